@@ -1,0 +1,107 @@
+"""Allocation-state persistence.
+
+Mirrors the reference's Storage interface {Save; Load; LoadOrCreate; Delete;
+ForEach; Close} over a host-persisted file (ref: pkg/storage/storage.go:15-93).
+The record format is identical — key ``namespace/name``, value = JSON
+container→Device map — but the container is SQLite in WAL mode instead of
+BoltDB: crash-safe, transactional, zero extra dependencies, and readable with
+stock tooling. ``boltcompat`` can read a reference agent's BoltDB file for
+one-shot migration (see ``migrate_from_bolt``).
+"""
+from __future__ import annotations
+
+import os
+import sqlite3
+import threading
+from typing import Callable, Optional
+
+from ..types import PodInfo
+
+
+class NotFoundError(KeyError):
+    pass
+
+
+class Storage:
+    """SQLite-backed pod-allocation store. Thread-safe."""
+
+    def __init__(self, db_path: str):
+        db_dir = os.path.dirname(os.path.abspath(db_path))
+        os.makedirs(db_dir, exist_ok=True)
+        self._path = db_path
+        self._lock = threading.Lock()
+        self._conn = sqlite3.connect(db_path, check_same_thread=False)
+        self._conn.execute("PRAGMA journal_mode=WAL")
+        self._conn.execute("PRAGMA synchronous=NORMAL")
+        self._conn.execute(
+            "CREATE TABLE IF NOT EXISTS pods (key TEXT PRIMARY KEY, val BLOB NOT NULL)"
+        )
+        self._conn.commit()
+
+    @property
+    def path(self) -> str:
+        return self._path
+
+    def save(self, pod_info: PodInfo) -> None:
+        with self._lock:
+            self._conn.execute(
+                "INSERT INTO pods(key, val) VALUES(?, ?) "
+                "ON CONFLICT(key) DO UPDATE SET val=excluded.val",
+                (pod_info.key(), pod_info.val()),
+            )
+            self._conn.commit()
+
+    def load(self, namespace: str, name: str) -> PodInfo:
+        key = f"{namespace}/{name}"
+        with self._lock:
+            row = self._conn.execute("SELECT val FROM pods WHERE key=?", (key,)).fetchone()
+        if row is None:
+            raise NotFoundError(key)
+        return PodInfo.from_raw(key, row[0])
+
+    def load_or_create(self, namespace: str, name: str) -> PodInfo:
+        try:
+            return self.load(namespace, name)
+        except NotFoundError:
+            return PodInfo(namespace=namespace, name=name)
+
+    def delete(self, namespace: str, name: str) -> None:
+        with self._lock:
+            self._conn.execute("DELETE FROM pods WHERE key=?", (f"{namespace}/{name}",))
+            self._conn.commit()
+
+    def for_each(self, fn: Callable[[PodInfo], None]) -> None:
+        with self._lock:
+            rows = self._conn.execute("SELECT key, val FROM pods").fetchall()
+        for key, val in rows:
+            fn(PodInfo.from_raw(key, val))
+
+    def close(self) -> None:
+        with self._lock:
+            self._conn.close()
+
+
+def migrate_from_bolt(bolt_path: str, storage: Storage) -> int:
+    """Import a reference agent's BoltDB state file. Returns records imported."""
+    from .boltcompat import read_bolt_bucket
+
+    n = 0
+    for key, val in read_bolt_bucket(bolt_path, b"root"):
+        storage.save(PodInfo.from_raw(key.decode(), val))
+        n += 1
+    return n
+
+
+def new_storage(db_path: str) -> Storage:
+    """Open the store; if ``db_path`` holds a BoltDB file from the reference
+    agent, migrate it in place (the original is kept with a ``.bolt-bak``
+    suffix)."""
+    from .boltcompat import is_bolt_file
+
+    if os.path.exists(db_path) and is_bolt_file(db_path):
+        bak = db_path + ".bolt-bak"
+        os.replace(db_path, bak)
+        st = Storage(db_path)
+        migrate_from_bolt(bak, st)
+        return st
+    return Storage(db_path)

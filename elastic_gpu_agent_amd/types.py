@@ -1,0 +1,119 @@
+"""Domain types: device-set identity and per-pod allocation records.
+
+Wire/disk compatible with the reference's state (ref: pkg/types/device.go:11-54,
+pkg/types/pod.go:10-62): a Device is a sorted fake-device-ID list identified by
+the first 8 hex chars of sha256(":".join(sorted_ids)); pod state serializes as
+JSON ``{container: {"Hash": .., "List": [..], "ResourceName": ..}}`` keyed by
+``namespace/name`` so a node can be migrated from the reference agent's DB.
+"""
+from __future__ import annotations
+
+import hashlib
+import json
+from dataclasses import dataclass, field
+from typing import Dict, Iterable, List, Optional
+
+
+def hash_device_ids(sorted_ids: List[str]) -> str:
+    return hashlib.sha256(":".join(sorted_ids).encode()).hexdigest()[:8]
+
+
+@dataclass(frozen=True)
+class Device:
+    """An allocated set of fake device IDs for one resource name."""
+
+    hash: str
+    list: tuple
+    resource_name: str
+
+    @staticmethod
+    def new(device_ids: Iterable[str], resource_name: str = "") -> "Device":
+        ids = tuple(sorted(device_ids))
+        return Device(hash=hash_device_ids(list(ids)), list=ids, resource_name=resource_name)
+
+    def equals(self, other: "Device") -> bool:
+        return (
+            self.hash == other.hash
+            and self.list == other.list
+            and self.resource_name == other.resource_name
+        )
+
+    # JSON shape uses Go field names for reference-state compatibility.
+    def to_json_obj(self) -> dict:
+        return {"Hash": self.hash, "List": list(self.list), "ResourceName": self.resource_name}
+
+    @staticmethod
+    def from_json_obj(obj: dict) -> "Device":
+        return Device(
+            hash=obj.get("Hash", ""),
+            list=tuple(obj.get("List") or ()),
+            resource_name=obj.get("ResourceName", ""),
+        )
+
+
+@dataclass(frozen=True)
+class PodContainer:
+    namespace: str
+    name: str
+    container: str
+
+    def __str__(self) -> str:
+        return f"{self.namespace}/{self.name}:{self.container}"
+
+    def pod(self) -> str:
+        return f"{self.namespace}/{self.name}"
+
+
+@dataclass
+class PodInfo:
+    namespace: str
+    name: str
+    container_device_map: Dict[str, Device] = field(default_factory=dict)
+
+    def key(self) -> str:
+        return f"{self.namespace}/{self.name}"
+
+    def val(self) -> bytes:
+        return json.dumps(
+            {c: d.to_json_obj() for c, d in self.container_device_map.items()},
+            separators=(",", ":"),
+        ).encode()
+
+    @staticmethod
+    def from_raw(key: str, val: bytes) -> "PodInfo":
+        parts = key.split("/")
+        if len(parts) != 2:
+            raise ValueError(f"error key format: {key}")
+        obj = json.loads(val.decode()) if val else {}
+        return PodInfo(
+            namespace=parts[0],
+            name=parts[1],
+            container_device_map={c: Device.from_json_obj(d) for c, d in obj.items()},
+        )
+
+
+@dataclass
+class GPUDevice:
+    """One enumerated physical MI355X (or fake) GPU.
+
+    The reference's operator returned {UUID, GPUIndex, Memory}
+    (ref: pkg/operator/base.go:77-83); the MI355X-native record additionally
+    carries everything placement and materialization need: the DRM render-node
+    minor (the injectable device node), CU count, NUMA node, and the xGMI peer
+    set used for topology-aware preferred allocation.
+    """
+
+    uuid: str
+    index: int
+    memory_bytes: int
+    drm_render_minor: int = 128
+    drm_card: int = 0
+    cu_count: int = 256
+    xcd_count: int = 8
+    numa_node: int = 0
+    xgmi_peers: tuple = ()  # GPU indexes directly linked over xGMI
+    compute_partition: str = "SPX"
+
+    @property
+    def memory_mib(self) -> int:
+        return self.memory_bytes // (1024 * 1024)

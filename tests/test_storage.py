@@ -1,0 +1,154 @@
+"""Storage tests: CRUD + reopen persistence + iteration + bolt migration.
+
+Covers what the reference's (stale, non-compiling) storage test intended
+(ref: pkg/storage/storage_test.go) plus the BoltDB migration path.
+"""
+import struct
+
+import pytest
+
+from elastic_gpu_agent_amd.storage import NotFoundError, Storage, migrate_from_bolt, new_storage
+from elastic_gpu_agent_amd.types import Device, PodInfo
+
+
+def make_pi(ns="ns", name="p1", container="c", ids=("0-00",)):
+    pi = PodInfo(namespace=ns, name=name)
+    pi.container_device_map[container] = Device.new(list(ids), "elasticgpu.io/gpu-core")
+    return pi
+
+
+def test_save_load_roundtrip_and_reopen(tmp_db):
+    st = Storage(tmp_db)
+    pi = make_pi()
+    st.save(pi)
+    got = st.load("ns", "p1")
+    assert got.container_device_map["c"].equals(pi.container_device_map["c"])
+    st.close()
+    # reopen: state survives restarts (the reference keeps its DB on the host)
+    st2 = Storage(tmp_db)
+    got2 = st2.load("ns", "p1")
+    assert got2.container_device_map["c"].hash == pi.container_device_map["c"].hash
+    st2.close()
+
+
+def test_load_miss_and_load_or_create(tmp_db):
+    st = Storage(tmp_db)
+    with pytest.raises(NotFoundError):
+        st.load("ns", "missing")
+    pi = st.load_or_create("ns", "missing")
+    assert pi.key() == "ns/missing" and pi.container_device_map == {}
+    st.close()
+
+
+def test_delete(tmp_db):
+    st = Storage(tmp_db)
+    st.save(make_pi())
+    st.delete("ns", "p1")
+    with pytest.raises(NotFoundError):
+        st.load("ns", "p1")
+    # deleting a non-existent key is a no-op
+    st.delete("ns", "p1")
+    st.close()
+
+
+def test_for_each(tmp_db):
+    st = Storage(tmp_db)
+    for i in range(5):
+        st.save(make_pi(name=f"p{i}"))
+    seen = []
+    st.for_each(lambda pi: seen.append(pi.key()))
+    assert sorted(seen) == [f"ns/p{i}" for i in range(5)]
+    st.close()
+
+
+def test_save_overwrites(tmp_db):
+    st = Storage(tmp_db)
+    st.save(make_pi(ids=("0-00",)))
+    st.save(make_pi(ids=("0-00", "0-01")))
+    got = st.load("ns", "p1")
+    assert len(got.container_device_map["c"].list) == 2
+    st.close()
+
+
+# ---- BoltDB migration ------------------------------------------------------
+
+def _synth_bolt_file(path, items, page_size=4096):
+    """Synthesize a minimal valid BoltDB file: 2 meta pages, a freelist page,
+    and one root-bucket leaf page holding bucket "root" INLINE with ``items``.
+
+    Built from the public Bolt format (magic 0xED0CDAED, version 2). This is a
+    writer implemented only for the test; the production code is read-only.
+    """
+    def page_header(pgid, flags, count, overflow=0):
+        return struct.pack("<QHHI", pgid, flags, count, overflow)
+
+    def leaf_page_body(kvs, bucket_flags=0):
+        # elements then key/value blobs; pos is relative to each element start
+        n = len(kvs)
+        elems = b""
+        blob = b""
+        elem_area = n * 16
+        for i, (k, v) in enumerate(kvs):
+            pos = (elem_area - i * 16) + len(blob)
+            elems += struct.pack("<IIII", bucket_flags, pos, len(k), len(v))
+            blob += k + v
+        return elems + blob, n
+
+    # inner bucket "root" as an inline bucket value:
+    inner_body, inner_n = leaf_page_body(items)
+    inline_page = page_header(0, 0x02, inner_n) + inner_body
+    bucket_val = struct.pack("<QQ", 0, 0) + inline_page  # root pgid 0 => inline
+
+    # page 3: root-bucket leaf with one bucket element
+    root_body, root_n = leaf_page_body([(b"root", bucket_val)], bucket_flags=0x01)
+    page3 = page_header(3, 0x02, root_n) + root_body
+
+    # page 2: empty freelist
+    page2 = page_header(2, 0x10, 0)
+
+    def meta(pgid, txid):
+        body = struct.pack(
+            "<IIII QQ QQQ Q",
+            0xED0CDAED, 2, page_size, 0,
+            3, 0,        # root bucket pgid=3, sequence
+            2, 4, txid,  # freelist pgid, high water, txid
+            0,           # checksum (unvalidated by our reader)
+        )
+        return page_header(pgid, 0x04, 0) + body
+
+    pages = [meta(0, 0), meta(1, 1), page2, page3]
+    with open(path, "wb") as f:
+        for p in pages:
+            assert len(p) <= page_size
+            f.write(p + b"\x00" * (page_size - len(p)))
+
+
+def test_bolt_migration(tmp_path):
+    bolt = str(tmp_path / "meta.db")
+    pi = make_pi(ns="default", name="bolt-pod")
+    _synth_bolt_file(bolt, [(pi.key().encode(), pi.val())])
+
+    from elastic_gpu_agent_amd.storage.boltcompat import is_bolt_file, read_bolt_bucket
+
+    assert is_bolt_file(bolt)
+    items = read_bolt_bucket(bolt, b"root")
+    assert items == [(b"default/bolt-pod", pi.val())]
+
+    # new_storage transparently migrates a bolt file found at db_path
+    st = new_storage(bolt)
+    got = st.load("default", "bolt-pod")
+    assert got.container_device_map["c"].equals(pi.container_device_map["c"])
+    st.close()
+
+
+def test_is_bolt_file_rejects_sqlite(tmp_db):
+    st = Storage(tmp_db)
+    st.save(make_pi())
+    st.close()
+    from elastic_gpu_agent_amd.storage.boltcompat import is_bolt_file
+
+    assert not is_bolt_file(tmp_db)
+    # new_storage on an existing sqlite file just opens it
+    st2 = new_storage(tmp_db)
+    assert st2.load("ns", "p1").name == "p1"
+    st2.close()

@@ -1,0 +1,80 @@
+"""Agent entrypoint (DaemonSet container).
+
+Flags mirror the reference's (ref: cmd/main.go:17-35) plus MI355X options.
+SIGUSR1 dumps all thread stacks (the reference's goroutine-dump equivalent,
+ref: pkg/common/util.go:58-66) via faulthandler.
+"""
+from __future__ import annotations
+
+import argparse
+import faulthandler
+import logging
+import signal
+import sys
+import threading
+
+
+def main(argv=None) -> int:
+    parser = argparse.ArgumentParser(prog="elastic-gpu-agent-amd")
+    parser.add_argument("--nodeName", "-nodeName", default="", help="node this agent runs on")
+    parser.add_argument(
+        "--dbFile", "-dbFile", default="/host/var/lib/egpu/meta.db", help="allocation state DB"
+    )
+    parser.add_argument("--kubeconf", "-kubeconf", default=None, help="kubeconfig path")
+    parser.add_argument(
+        "--gpuPluginName", "-gpuPluginName", default="gpushare", help="plugin to run"
+    )
+    parser.add_argument(
+        "--backend", default="amdsmi", choices=["amdsmi", "fake"],
+        help="GPU enumeration backend (fake = synthetic gfx950 fleet)",
+    )
+    parser.add_argument(
+        "--mem-unit-mib", type=int, default=1,
+        help="MiB of HBM per gpu-memory unit (1 = reference contract; "
+        "1024 recommended for 288 GB parts if kubelet device counts bite)",
+    )
+    parser.add_argument(
+        "--no-isolation", action="store_true",
+        help="disable HSA-shim CU-mask/HBM-quota injection",
+    )
+    parser.add_argument("--metrics-port", type=int, default=0, help="Prometheus port (0=off)")
+    parser.add_argument("-v", "--verbose", action="count", default=0)
+    args = parser.parse_args(argv)
+
+    logging.basicConfig(
+        level=logging.DEBUG if args.verbose else logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s: %(message)s",
+    )
+    faulthandler.register(signal.SIGUSR1, all_threads=True)
+
+    from ..manager import GPUManager, ManagerOptions
+    from ..plugins.config import PluginOptions
+
+    opts = ManagerOptions(
+        node_name=args.nodeName,
+        db_path=args.dbFile,
+        kubeconf=args.kubeconf,
+        gpu_plugin_name=args.gpuPluginName,
+        backend=args.backend,
+        plugin_options=PluginOptions(
+            mem_unit_mib=args.mem_unit_mib, isolation=not args.no_isolation
+        ),
+        metrics_port=args.metrics_port,
+    )
+    manager = GPUManager(opts)
+    manager.run()
+
+    stop = threading.Event()
+
+    def _term(signum, frame):
+        stop.set()
+
+    signal.signal(signal.SIGTERM, _term)
+    signal.signal(signal.SIGINT, _term)
+    stop.wait()
+    manager.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,171 @@
+"""Isolation control plane: CU-mask bookkeeping + per-allocation limits files.
+
+The data plane is the hand-written HSA interposer (native/egpu_shim.cpp,
+loaded into containers via HSA_TOOLS_LIB) which applies CU masks with
+``hsa_amd_queue_cu_set_mask`` and enforces HBM quotas by intercepting HSA
+memory-pool allocation. This module is the agent-side half:
+
+- ``CUMaskAllocator``: picks disjoint XCD-round-robin CU sets per GPU for
+  live fractional allocations (persisted in the storage aux table so masks
+  survive agent restarts and are reclaimed by GC).
+- ``LimitsWriter``: writes the per-allocation ``<hash>.json`` limits file the
+  shim reads inside the container (mounted by the Allocate response).
+
+The reference has no open-source equivalent (its QoS lived in the closed qGPU
+driver) — this layer is MI355X-scoped, not a port.
+"""
+from __future__ import annotations
+
+import json
+import os
+import threading
+from typing import Dict, List, Optional, Tuple
+
+from .. import consts
+from ..types import GPUDevice
+from .cumask import mask_hex, mask_words_from_cus, parse_mask_hex, popcount
+
+AUX_MASK_PREFIX = "mask/"  # aux key: mask/<alloc_hash> -> json record
+
+
+class CUMaskAllocator:
+    """Allocates disjoint CU sets (spread across XCDs) per physical GPU.
+
+    Occupancy is reconstructed from the storage aux table on startup, so a
+    restarted agent keeps honoring masks of running pods (the Restore path
+    the reference declared but never implemented — pkg/manager/manager.go:20).
+    """
+
+    def __init__(self, storage, devices: List[GPUDevice]):
+        self._storage = storage
+        self._devices = {d.index: d for d in devices}
+        self._lock = threading.Lock()
+
+    # ---- occupancy ----
+    def _live_cus(self, gpu_index: int) -> set:
+        used = set()
+        for key, val in self._storage.aux_items(AUX_MASK_PREFIX):
+            rec = json.loads(val)
+            if rec.get("gpu_index") == gpu_index:
+                words = parse_mask_hex(rec["cu_mask"])
+                for w_i, w in enumerate(words):
+                    for b in range(32):
+                        if w >> b & 1:
+                            used.add(w_i * 32 + b)
+        return used
+
+    def allocate(self, alloc_hash: str, gpu_index: int, percent: int) -> Tuple[str, int]:
+        """Pick a CU set of ``percent``% of the GPU, disjoint from live
+        allocations when capacity allows (oversubscription falls back to
+        overlapping masks — documented QoS mode). Returns (mask_hex, n_cus)."""
+        dev = self._devices.get(gpu_index)
+        total = dev.cu_count if dev else consts.GFX950_CU_COUNT
+        xcds = dev.xcd_count if dev else consts.GFX950_XCD_COUNT
+        per_xcd = total // xcds
+        from .cumask import cu_count_for_percent
+
+        n = cu_count_for_percent(percent, total)
+        with self._lock:
+            used = self._live_cus(gpu_index)
+            base, extra = divmod(n, xcds)
+            cus: List[int] = []
+            for xcd in range(xcds):
+                want = base + (1 if xcd < extra else 0)
+                free = [
+                    xcd * per_xcd + k for k in range(per_xcd) if xcd * per_xcd + k not in used
+                ]
+                take = free[:want]
+                if len(take) < want:
+                    # oversubscribed: wrap around, overlapping least-recently
+                    # chosen CUs of this XCD
+                    all_cus = [xcd * per_xcd + k for k in range(per_xcd)]
+                    for c in all_cus:
+                        if len(take) >= want:
+                            break
+                        if c not in take:
+                            take.append(c)
+                cus.extend(take)
+            words = mask_words_from_cus(cus, total)
+            hexmask = mask_hex(words)
+            self._storage.aux_set(
+                AUX_MASK_PREFIX + alloc_hash,
+                json.dumps(
+                    {"gpu_index": gpu_index, "cu_mask": hexmask, "cu_count": n, "percent": percent}
+                ),
+            )
+        return hexmask, n
+
+    def release(self, alloc_hash: str) -> None:
+        self._storage.aux_delete(AUX_MASK_PREFIX + alloc_hash)
+
+    def get(self, alloc_hash: str) -> Optional[dict]:
+        raw = self._storage.aux_get(AUX_MASK_PREFIX + alloc_hash)
+        return json.loads(raw) if raw else None
+
+
+class LimitsWriter:
+    """Per-allocation limits files consumed by the HSA shim in-container.
+
+    Layout: ``<limits_dir>/<hash>.json`` on the host, mounted read-only at
+    ``/etc/egpu/limits-<kind>.json`` in the container. Written (empty) at
+    Allocate time so kubelet can mount it, finalized at PreStart once the
+    physical GPU binding is known."""
+
+    def __init__(self, limits_dir: str):
+        self.limits_dir = limits_dir
+        os.makedirs(limits_dir, exist_ok=True)
+
+    def host_path(self, alloc_hash: str) -> str:
+        return os.path.join(self.limits_dir, f"{alloc_hash}.json")
+
+    @staticmethod
+    def container_path(kind: str) -> str:
+        return f"/etc/egpu/limits-{kind}.json"
+
+    def touch(self, alloc_hash: str) -> str:
+        p = self.host_path(alloc_hash)
+        if not os.path.exists(p):
+            self._atomic_write(p, {})
+        return p
+
+    def finalize(
+        self,
+        alloc_hash: str,
+        gpu_indexes: List[int],
+        devices: List[GPUDevice],
+        cu_mask: Optional[str] = None,
+        cu_count: Optional[int] = None,
+        mem_limit_bytes: Optional[int] = None,
+    ) -> None:
+        dev_by_idx = {d.index: d for d in devices}
+        rec: Dict = {
+            "version": 1,
+            "gpu_indexes": gpu_indexes,
+            "render_minors": [
+                dev_by_idx[i].drm_render_minor for i in gpu_indexes if i in dev_by_idx
+            ],
+            "uuids": [dev_by_idx[i].uuid for i in gpu_indexes if i in dev_by_idx],
+        }
+        if cu_mask is not None:
+            rec["cu_mask"] = cu_mask
+            rec["cu_count"] = cu_count
+        if mem_limit_bytes is not None:
+            rec["mem_limit_bytes"] = mem_limit_bytes
+        self._atomic_write(self.host_path(alloc_hash), rec)
+
+    def delete(self, alloc_hash: str) -> None:
+        try:
+            os.unlink(self.host_path(alloc_hash))
+        except FileNotFoundError:
+            pass
+
+    def read(self, alloc_hash: str) -> dict:
+        with open(self.host_path(alloc_hash)) as f:
+            return json.load(f)
+
+    @staticmethod
+    def _atomic_write(path: str, obj: dict) -> None:
+        tmp = path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(obj, f)
+        os.replace(tmp, path)

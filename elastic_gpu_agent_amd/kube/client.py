@@ -1,0 +1,114 @@
+"""Minimal Kubernetes API client (list/watch/get pods for one node).
+
+The image has no `kubernetes` package, so this speaks the REST API directly
+with httpx: in-cluster service-account auth (token + CA from
+/var/run/secrets/kubernetes.io/serviceaccount) or a kubeconfig file.
+Only what the agent needs (ref behavior: pkg/kube/sitter.go:42-48,58-71 —
+node-filtered pod list/watch + direct gets).
+"""
+from __future__ import annotations
+
+import json
+import os
+import threading
+from typing import Callable, Iterator, List, Optional
+
+import httpx
+
+from .. import consts
+from .pods import Pod
+
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
+class NotFound(Exception):
+    pass
+
+
+class K8sClient:
+    def __init__(
+        self,
+        base_url: Optional[str] = None,
+        token: Optional[str] = None,
+        verify=None,
+        kubeconf: Optional[str] = None,
+    ):
+        if kubeconf:
+            base_url, token, verify = self._from_kubeconfig(kubeconf)
+        elif base_url is None:
+            host = os.environ.get("KUBERNETES_SERVICE_HOST")
+            port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+            if not host:
+                raise RuntimeError("not in cluster and no kubeconf given")
+            base_url = f"https://{host}:{port}"
+            with open(os.path.join(SA_DIR, "token")) as f:
+                token = f.read().strip()
+            ca = os.path.join(SA_DIR, "ca.crt")
+            verify = ca if os.path.exists(ca) else False
+        headers = {}
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        self._client = httpx.Client(
+            base_url=base_url, headers=headers, verify=verify if verify is not None else False,
+            timeout=30.0,
+        )
+
+    @staticmethod
+    def _from_kubeconfig(path: str):
+        import yaml
+
+        with open(path) as f:
+            cfg = yaml.safe_load(f)
+        ctx_name = cfg.get("current-context")
+        ctx = next(c["context"] for c in cfg["contexts"] if c["name"] == ctx_name)
+        cluster = next(c["cluster"] for c in cfg["clusters"] if c["name"] == ctx["cluster"])
+        user = next(u["user"] for u in cfg["users"] if u["name"] == ctx["user"])
+        token = user.get("token")
+        verify: object = cluster.get("certificate-authority", False)
+        if cluster.get("insecure-skip-tls-verify"):
+            verify = False
+        return cluster["server"], token, verify
+
+    # ---- pods ----
+    def list_pods(self, node_name: str) -> tuple:
+        """Returns (pods, resource_version)."""
+        r = self._client.get(
+            "/api/v1/pods",
+            params={"fieldSelector": f"{consts.NODE_NAME_FIELD}={node_name}"},
+        )
+        r.raise_for_status()
+        obj = r.json()
+        rv = obj.get("metadata", {}).get("resourceVersion", "")
+        return [Pod.from_api_obj(p) for p in obj.get("items", [])], rv
+
+    def watch_pods(self, node_name: str, resource_version: str) -> Iterator[tuple]:
+        """Yields (event_type, Pod). Terminates on stream end/error."""
+        with self._client.stream(
+            "GET",
+            "/api/v1/pods",
+            params={
+                "watch": "true",
+                "fieldSelector": f"{consts.NODE_NAME_FIELD}={node_name}",
+                "resourceVersion": resource_version,
+                "allowWatchBookmarks": "true",
+            },
+            timeout=httpx.Timeout(30.0, read=None),
+        ) as resp:
+            resp.raise_for_status()
+            for line in resp.iter_lines():
+                if not line:
+                    continue
+                ev = json.loads(line)
+                if ev.get("type") == "BOOKMARK":
+                    continue
+                yield ev.get("type", ""), Pod.from_api_obj(ev.get("object", {}))
+
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        r = self._client.get(f"/api/v1/namespaces/{namespace}/pods/{name}")
+        if r.status_code == 404:
+            raise NotFound(f"{namespace}/{name}")
+        r.raise_for_status()
+        return Pod.from_api_obj(r.json())
+
+    def close(self):
+        self._client.close()

@@ -1,0 +1,112 @@
+"""Device locator: fake-device-ID set → {namespace, pod, container}.
+
+Asks the kubelet podresources service which pod/container was assigned a set
+of device IDs (ref: pkg/kube/locator.go:18-118). Matching is uniform across
+kubelet versions: per container, the IDs of all ContainerDevices entries for
+our resource are merged before hashing, which covers both the ≤1.20 shape
+(all IDs in one entry) and the ≥1.21 shape (one ID per entry) without a dual
+code path.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+from typing import List, Optional
+
+import grpc
+
+from .. import consts
+from ..protos import podresources as pr
+from ..types import Device, PodContainer
+
+log = logging.getLogger(__name__)
+
+
+class DeviceLocator:
+    def locate(self, device: Device) -> PodContainer:
+        raise NotImplementedError
+
+    def close(self) -> None:
+        pass
+
+
+class KubeletDeviceLocator(DeviceLocator):
+    def __init__(self, resource_name: str, socket_path: str = consts.POD_RESOURCES_SOCKET):
+        self._resource = resource_name
+        self._socket = socket_path
+        self._lock = threading.Lock()
+        self._channel: Optional[grpc.Channel] = None
+        self._list = None
+
+    def _ensure(self):
+        if self._channel is None:
+            self._channel = grpc.insecure_channel(
+                f"unix://{self._socket}",
+                options=[
+                    ("grpc.max_receive_message_length", consts.POD_RESOURCES_MAX_SIZE),
+                    ("grpc.max_send_message_length", consts.POD_RESOURCES_MAX_SIZE),
+                ],
+            )
+            self._list = self._channel.unary_unary(
+                pr.METHOD_LIST,
+                request_serializer=pr.ListPodResourcesRequest.encode,
+                response_deserializer=pr.ListPodResourcesResponse.decode,
+            )
+
+    def _reset(self):
+        if self._channel is not None:
+            self._channel.close()
+        self._channel = None
+        self._list = None
+
+    def list_once(self) -> dict:
+        with self._lock:
+            self._ensure()
+            try:
+                return self._list({}, timeout=10.0)
+            except grpc.RpcError:
+                # lazy repair: re-dial once (kubelet may have restarted,
+                # ref behavior: pkg/kube/locator.go:47-53)
+                self._reset()
+                self._ensure()
+                return self._list({}, timeout=10.0)
+
+    def locate(self, device: Device) -> PodContainer:
+        resp = self.list_once()
+        for pod in resp.get("pod_resources", []):
+            for container in pod.get("containers", []):
+                ids: List[str] = []
+                for cd in container.get("devices", []):
+                    if cd.get("resource_name") == self._resource:
+                        ids.extend(cd.get("device_ids", []))
+                if not ids:
+                    continue
+                if Device.new(ids).hash == device.hash:
+                    return PodContainer(
+                        namespace=pod.get("namespace", ""),
+                        name=pod.get("name", ""),
+                        container=container.get("name", ""),
+                    )
+        raise KeyError(
+            f"no pod/container holds device set {device.hash} of {self._resource}"
+        )
+
+    def close(self) -> None:
+        with self._lock:
+            self._reset()
+
+
+class FakeDeviceLocator(DeviceLocator):
+    """Test/bench locator: direct hash→PodContainer table."""
+
+    def __init__(self):
+        self.table = {}
+
+    def assign(self, device_hash: str, pc: PodContainer) -> None:
+        self.table[device_hash] = pc
+
+    def locate(self, device: Device) -> PodContainer:
+        pc = self.table.get(device.hash)
+        if pc is None:
+            raise KeyError(f"no assignment for {device.hash}")
+        return pc

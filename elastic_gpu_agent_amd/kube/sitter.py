@@ -1,0 +1,149 @@
+"""Pod sitter: node-filtered pod cache with delete hooks.
+
+Mirrors the reference Sitter interface {Start; GetPod; GetPodFromApiServer}
+plus a pod-delete hook feeding GC (ref: pkg/kube/sitter.go:18-77). The real
+implementation runs a list+watch loop against the API server; FakeSitter
+backs tests and the benchmark driver.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Callable, Dict, Optional
+
+from .client import K8sClient, NotFound
+from .pods import Pod
+
+log = logging.getLogger(__name__)
+
+
+class Sitter:
+    """Interface: start(), has_synced(), get_pod(), get_pod_from_api_server()."""
+
+    def start(self) -> None:
+        raise NotImplementedError
+
+    def has_synced(self) -> bool:
+        raise NotImplementedError
+
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        raise NotImplementedError
+
+    def get_pod_from_api_server(self, namespace: str, name: str) -> Pod:
+        raise NotImplementedError
+
+    def stop(self) -> None:
+        pass
+
+
+class PodSitter(Sitter):
+    def __init__(
+        self,
+        client: K8sClient,
+        node_name: str,
+        delete_hook: Optional[Callable[[Pod], None]] = None,
+        relist_interval: float = 30.0,
+    ):
+        self._client = client
+        self._node = node_name
+        self._hook = delete_hook
+        self._relist = relist_interval
+        self._cache: Dict[str, Pod] = {}
+        self._lock = threading.Lock()
+        self._synced = threading.Event()
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> None:
+        self._thread = threading.Thread(target=self._run, name="pod-sitter", daemon=True)
+        self._thread.start()
+
+    def _run(self) -> None:
+        while not self._stop.is_set():
+            try:
+                pods, rv = self._client.list_pods(self._node)
+                with self._lock:
+                    old = set(self._cache)
+                    self._cache = {p.namespace + "/" + p.name: p for p in pods}
+                    gone = old - set(self._cache)
+                self._synced.set()
+                if self._hook:
+                    for key in gone:
+                        ns, name = key.split("/", 1)
+                        self._hook(Pod(namespace=ns, name=name, deleted=True))
+                for ev_type, pod in self._client.watch_pods(self._node, rv):
+                    if self._stop.is_set():
+                        return
+                    key = pod.namespace + "/" + pod.name
+                    if ev_type == "DELETED":
+                        with self._lock:
+                            self._cache.pop(key, None)
+                        if self._hook:
+                            self._hook(pod)
+                    else:
+                        with self._lock:
+                            self._cache[key] = pod
+            except Exception as e:  # relist after transient API errors
+                log.warning("sitter list/watch error: %s; relisting in 1s", e)
+                time.sleep(1.0)
+
+    def has_synced(self) -> bool:
+        return self._synced.is_set()
+
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        with self._lock:
+            pod = self._cache.get(f"{namespace}/{name}")
+        if pod is None:
+            raise NotFound(f"{namespace}/{name}")
+        return pod
+
+    def get_pod_from_api_server(self, namespace: str, name: str) -> Pod:
+        return self._client.get_pod(namespace, name)
+
+    def stop(self) -> None:
+        self._stop.set()
+
+
+class FakeSitter(Sitter):
+    """Dict-backed sitter for tests and the CPU benchmark configs."""
+
+    def __init__(self):
+        self.pods: Dict[str, Pod] = {}
+        self.api_pods: Dict[str, Pod] = {}
+        self._hook: Optional[Callable[[Pod], None]] = None
+
+    def set_delete_hook(self, hook: Callable[[Pod], None]) -> None:
+        self._hook = hook
+
+    def add(self, pod: Pod, api: bool = True) -> None:
+        key = f"{pod.namespace}/{pod.name}"
+        self.pods[key] = pod
+        if api:
+            self.api_pods[key] = pod
+
+    def remove(self, namespace: str, name: str) -> None:
+        key = f"{namespace}/{name}"
+        pod = self.pods.pop(key, None)
+        self.api_pods.pop(key, None)
+        if pod and self._hook:
+            pod.deleted = True
+            self._hook(pod)
+
+    def start(self) -> None:
+        pass
+
+    def has_synced(self) -> bool:
+        return True
+
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        pod = self.pods.get(f"{namespace}/{name}")
+        if pod is None:
+            raise NotFound(f"{namespace}/{name}")
+        return pod
+
+    def get_pod_from_api_server(self, namespace: str, name: str) -> Pod:
+        pod = self.api_pods.get(f"{namespace}/{name}")
+        if pod is None:
+            raise NotFound(f"{namespace}/{name}")
+        return pod

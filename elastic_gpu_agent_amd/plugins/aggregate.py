@@ -1,0 +1,142 @@
+"""GPUShare plugin aggregate: both resource servers + GC + Restore.
+
+Role of the reference's GPUSharePlugin (ref: pkg/plugins/base.go:203-306)
+with two additions: GC also reclaims CU masks and limits files, and
+Restore() — declared but never implemented in the reference
+(pkg/manager/manager.go:20) — re-creates per-allocation device links from
+the persisted state after a node reboot wiped /host/dev.
+"""
+from __future__ import annotations
+
+import logging
+import queue
+import threading
+import time
+from typing import List, Optional
+
+from .. import consts
+from ..kube.client import NotFound
+from ..types import PodInfo
+from .base import DevicePluginServer
+from .config import GPUPluginConfig
+from .gpushare import GPUShareCorePlugin, GPUShareMemoryPlugin
+
+log = logging.getLogger(__name__)
+
+GC_PERIOD_SECONDS = 60.0
+
+
+class GPUSharePlugin:
+    def __init__(self, config: GPUPluginConfig):
+        self.cfg = config
+        self.core = GPUShareCorePlugin(config)
+        self.memory = GPUShareMemoryPlugin(config)
+        self.core_server = DevicePluginServer(
+            self.core,
+            consts.RESOURCE_GPU_CORE,
+            consts.CORE_SOCK_NAME,
+            plugin_dir=config.paths.plugin_dir,
+            kubelet_socket=config.paths.kubelet_socket,
+        )
+        self.memory_server = DevicePluginServer(
+            self.memory,
+            consts.RESOURCE_GPU_MEMORY,
+            consts.MEMORY_SOCK_NAME,
+            plugin_dir=config.paths.plugin_dir,
+            kubelet_socket=config.paths.kubelet_socket,
+        )
+        self._stop = threading.Event()
+
+    # ---- lifecycle ----
+    def run(self) -> None:
+        self.core_server.start()
+        self.memory_server.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        self.core_server.stop()
+        self.memory_server.stop()
+
+    # ---- GC ----
+    def gc_once(self) -> int:
+        """Reconcile storage against live pods; returns records reclaimed."""
+        doomed: List[PodInfo] = []
+
+        def visit(pi: PodInfo):
+            try:
+                self.cfg.sitter.get_pod(pi.namespace, pi.name)
+                return
+            except NotFound:
+                pass
+            try:
+                self.cfg.sitter.get_pod_from_api_server(pi.namespace, pi.name)
+                return
+            except NotFound:
+                doomed.append(pi)
+            except Exception as e:
+                log.warning("GC: API check failed for %s/%s: %s", pi.namespace, pi.name, e)
+
+        self.cfg.storage.for_each(visit)
+        for pi in doomed:
+            for container, device in pi.container_device_map.items():
+                links = (
+                    GPUShareCorePlugin.links_for(len(device.list))
+                    if device.resource_name == consts.RESOURCE_GPU_CORE
+                    else 1
+                )
+                for i in range(links):
+                    self.cfg.operator.delete(-1, f"{device.hash}-{i}")
+                if self.cfg.cumask:
+                    self.cfg.cumask.release(device.hash)
+                if self.cfg.limits:
+                    self.cfg.limits.delete(device.hash)
+            self.cfg.storage.delete(pi.namespace, pi.name)
+            log.info("GC reclaimed %s/%s", pi.namespace, pi.name)
+        return len(doomed)
+
+    def gc_loop(self, gc_events: "queue.Queue", period: float = GC_PERIOD_SECONDS) -> None:
+        """Event-driven + periodic reconciliation (ref: pkg/plugins/base.go:241-306)."""
+        while not self._stop.is_set():
+            try:
+                gc_events.get(timeout=period)
+            except queue.Empty:
+                pass
+            try:
+                self.gc_once()
+            except Exception as e:
+                log.error("GC pass failed: %s", e)
+
+    # ---- Restore ----
+    def restore(self) -> int:
+        """Re-create device links for pods that still exist (node reboot path).
+        Returns number of allocations restored."""
+        restored = 0
+        records: List[PodInfo] = []
+        self.cfg.storage.for_each(records.append)
+        for pi in records:
+            try:
+                pod = self.cfg.sitter.get_pod(pi.namespace, pi.name)
+            except NotFound:
+                try:
+                    pod = self.cfg.sitter.get_pod_from_api_server(pi.namespace, pi.name)
+                except Exception:
+                    continue  # GC will reclaim
+            for container, device in pi.container_device_map.items():
+                raw = pod.container_gpu_indexes(container)
+                if raw is None:
+                    continue
+                try:
+                    indexes = [int(x) for x in raw.split(",") if x != ""]
+                except ValueError:
+                    continue
+                links = (
+                    GPUShareCorePlugin.links_for(len(device.list))
+                    if device.resource_name == consts.RESOURCE_GPU_CORE
+                    else 1
+                )
+                for i, idx in enumerate(indexes[:links]):
+                    alloc_id = f"{device.hash}-{i}"
+                    if not self.cfg.operator.check(idx, alloc_id):
+                        self.cfg.operator.create(idx, alloc_id)
+                        restored += 1
+        return restored

@@ -1,0 +1,43 @@
+"""Shared dependency bundle for the device plugins
+(role of the reference's GPUPluginConfig, ref: pkg/plugins/base.go:32-50)."""
+from __future__ import annotations
+
+import dataclasses
+from typing import Optional
+
+from .. import consts
+
+
+@dataclasses.dataclass
+class AgentPaths:
+    """Host-side paths the agent writes / containers consume."""
+
+    dev_root: str = consts.HOST_DEV_ROOT  # where per-alloc symlinks live (host /dev)
+    plugin_dir: str = consts.DEVICE_PLUGIN_PATH
+    kubelet_socket: Optional[str] = None  # default: <plugin_dir>/kubelet.sock
+    podresources_socket: str = consts.POD_RESOURCES_SOCKET
+    limits_dir: str = "/host/var/lib/egpu/limits"
+    # host path of the HSA shim library mounted into containers; None disables
+    # isolation env injection entirely.
+    shim_host_path: Optional[str] = "/host/opt/egpu/libegpu_shim.so"
+    shim_container_path: str = "/opt/egpu/libegpu_shim.so"
+
+
+@dataclasses.dataclass
+class PluginOptions:
+    mem_unit_mib: int = 1  # MiB per gpu-memory unit (reference contract: 1)
+    isolation: bool = True  # inject HSA shim + limits into fractional pods
+    health_refresh_seconds: float = 30.0
+
+
+@dataclasses.dataclass
+class GPUPluginConfig:
+    operator: object  # operator.GPUOperator
+    storage: object  # storage.Storage
+    sitter: object  # kube.sitter.Sitter
+    core_locator: object  # kube.locator.DeviceLocator
+    memory_locator: object  # kube.locator.DeviceLocator
+    paths: AgentPaths = dataclasses.field(default_factory=AgentPaths)
+    options: PluginOptions = dataclasses.field(default_factory=PluginOptions)
+    limits: Optional[object] = None  # isolation.LimitsWriter
+    cumask: Optional[object] = None  # isolation.CUMaskAllocator

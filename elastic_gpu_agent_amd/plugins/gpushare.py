@@ -1,0 +1,312 @@
+"""GPUShare device plugins: `elasticgpu.io/gpu-core` and `gpu-memory`.
+
+Re-design of the reference's two gRPC servers (ref: pkg/plugins/gpushare.go)
+with the MI355X-native differences:
+
+- fake devices carry kubelet TopologyInfo (NUMA node of the GPU);
+- Allocate answers **per container** (the reference merged all containers
+  into one response — SURVEY §3.3 quirk, not copied);
+- Allocate's DeviceSpecs grant /dev/kfd plus the per-allocation symlink
+  (whose resolution by the container runtime whitelists the underlying
+  /dev/dri/renderD* major:minor in the device cgroup);
+- fractional core allocations mount the HSA shim + a limits file and set
+  HSA_TOOLS_LIB so CU masks / HBM quotas are enforced in-container;
+- GetPreferredAllocation is implemented (xGMI/NUMA-aware packing) where the
+  reference stubs it (pkg/plugins/base.go:94-96).
+"""
+from __future__ import annotations
+
+import logging
+import math
+import threading
+import time
+from typing import Dict, List, Optional
+
+from .. import consts, topology
+from ..kube.client import NotFound
+from ..types import Device, PodInfo
+from .config import GPUPluginConfig
+
+log = logging.getLogger(__name__)
+
+
+def _rpc_error(context, code, msg):
+    if context is not None:
+        import grpc
+
+        context.abort(code, msg)
+    raise RuntimeError(msg)
+
+
+class GPUSharePluginBase:
+    """Handlers shared by the core and memory resource servers."""
+
+    resource_name: str = ""
+
+    def __init__(self, config: GPUPluginConfig):
+        self.cfg = config
+        self._devices_lock = threading.Lock()
+        self._refresh = threading.Event()
+
+    # ---- device advertisement ----
+    def fake_device_ids_for_gpu(self, gpu) -> List[str]:
+        raise NotImplementedError
+
+    def list_devices(self) -> List[dict]:
+        out = []
+        for gpu in self.cfg.operator.devices():
+            topo = {"nodes": [{"ID": gpu.numa_node}]}
+            for did in self.fake_device_ids_for_gpu(gpu):
+                out.append({"ID": did, "health": consts.HEALTHY, "topology": topo})
+        return out
+
+    # ---- gRPC handlers ----
+    def get_device_plugin_options(self, request, context) -> dict:
+        return {"pre_start_required": True, "get_preferred_allocation_available": True}
+
+    def list_and_watch(self, context):
+        """Initial device list + re-advertisement when enumeration changes.
+
+        Unlike the reference (single static send, devices never re-checked —
+        SURVEY §3.2), the backend is re-enumerated periodically so a GPU
+        falling off the bus transitions its fake devices out of the list."""
+        current = self.list_devices()
+        yield {"devices": current}
+        interval = self.cfg.options.health_refresh_seconds
+        while context is None or context.is_active():
+            triggered = self._refresh.wait(timeout=interval)
+            self._refresh.clear()
+            if context is not None and not context.is_active():
+                return
+            try:
+                fresh_gpus = self.cfg.operator.devices(refresh=True)
+            except Exception as e:
+                log.error("device re-enumeration failed: %s", e)
+                continue
+            fresh = self.list_devices()
+            if fresh != current:
+                current = fresh
+                yield {"devices": current}
+            if triggered:
+                yield {"devices": current}
+
+    def trigger_refresh(self) -> None:
+        self._refresh.set()
+
+    def get_preferred_allocation(self, request, context) -> dict:
+        devices = self.cfg.operator.devices()
+        responses = []
+        for cr in request.get("container_requests", []):
+            picked = topology.prefer_allocation(
+                cr.get("available_deviceIDs", []),
+                cr.get("must_include_deviceIDs", []),
+                cr.get("allocation_size", 0),
+                devices,
+            )
+            responses.append({"deviceIDs": picked})
+        return {"container_responses": responses}
+
+    def allocate(self, request, context) -> dict:
+        responses = []
+        for cr in request.get("container_requests", []):
+            ids = cr.get("devicesIDs", [])
+            device = Device.new(ids, self.resource_name)
+            responses.append(self._allocate_one(device, ids))
+        return {"container_responses": responses}
+
+    def _allocate_one(self, device: Device, ids: List[str]) -> dict:
+        raise NotImplementedError
+
+    def _isolation_payload(self, device: Device, kind: str) -> dict:
+        """Common shim env + mounts for a fractional allocation."""
+        paths = self.cfg.paths
+        resp: dict = {"envs": {}, "mounts": []}
+        if not (self.cfg.options.isolation and self.cfg.limits and paths.shim_host_path):
+            return resp
+        limits_host = self.cfg.limits.touch(device.hash)
+        resp["mounts"] = [
+            {
+                "container_path": paths.shim_container_path,
+                "host_path": paths.shim_host_path,
+                "read_only": True,
+            },
+            {
+                "container_path": self.cfg.limits.container_path(kind),
+                "host_path": limits_host,
+                "read_only": True,
+            },
+        ]
+        resp["envs"] = {"HSA_TOOLS_LIB": paths.shim_container_path}
+        return resp
+
+    def pre_start_container(self, request, context) -> dict:
+        ids = request.get("devicesIDs", [])
+        device = Device.new(ids, self.resource_name)
+        try:
+            pc = self._locator().locate(device)
+        except KeyError as e:
+            return self._fail(context, f"locate {device.hash}: {e}")
+        try:
+            pod = self.cfg.sitter.get_pod(pc.namespace, pc.name)
+        except NotFound:
+            try:
+                pod = self.cfg.sitter.get_pod_from_api_server(pc.namespace, pc.name)
+            except NotFound:
+                return self._fail(context, f"pod {pc.pod()} not found")
+        if not pod.is_assumed():
+            return self._fail(
+                context, f"pod {pc.pod()} lacks {consts.ELASTIC_GPU_ASSUMED_ANNOTATION}=true"
+            )
+        raw = pod.container_gpu_indexes(pc.container)
+        if raw is None:
+            return self._fail(
+                context,
+                f"pod {pc.pod()} lacks annotation for container {pc.container}",
+            )
+        try:
+            indexes = [int(x) for x in raw.split(",") if x != ""]
+        except ValueError:
+            return self._fail(context, f"bad GPU index annotation {raw!r}")
+        created: List[str] = []
+        try:
+            self._bind(device, ids, indexes, created)
+        except Exception as e:
+            for alloc_id in created:  # rollback partial symlinks
+                self.cfg.operator.delete(-1, alloc_id)
+            return self._fail(context, f"bind {device.hash}: {e}")
+        pi = self.cfg.storage.load_or_create(pc.namespace, pc.name)
+        pi.container_device_map[pc.container] = device
+        self.cfg.storage.save(pi)
+        return {}
+
+    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str]):
+        raise NotImplementedError
+
+    def _locator(self):
+        raise NotImplementedError
+
+    @staticmethod
+    def _fail(context, msg):
+        log.error("%s", msg)
+        if context is not None:
+            import grpc
+
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, msg)
+        raise RuntimeError(msg)
+
+
+class GPUShareCorePlugin(GPUSharePluginBase):
+    """100 percent-unit fake devices per GPU; fractional pods get CU masks."""
+
+    resource_name = consts.RESOURCE_GPU_CORE
+
+    def fake_device_ids_for_gpu(self, gpu) -> List[str]:
+        return [f"{gpu.index}-{slot:02d}" for slot in range(consts.GPU_PERCENT_EACH_CARD)]
+
+    @staticmethod
+    def links_for(ids_count: int) -> int:
+        """Number of per-allocation GPU links: one per started 100 units."""
+        return max(1, math.ceil(ids_count / consts.GPU_PERCENT_EACH_CARD))
+
+    def _allocate_one(self, device: Device, ids: List[str]) -> dict:
+        n_links = self.links_for(len(ids))
+        devices_spec = [
+            {"container_path": consts.KFD_PATH, "host_path": consts.KFD_PATH,
+             "permissions": "rw"}
+        ]
+        for i in range(n_links):
+            host = f"/dev/{consts.ELASTIC_GPU_LINK_FMT % (device.hash + '-' + str(i))}"
+            devices_spec.append(
+                {
+                    "container_path": f"/dev/egpu/gpu{i}",
+                    "host_path": host,
+                    "permissions": "rw",
+                }
+            )
+        resp = {
+            "envs": {consts.GPU_ENV_KEY: device.hash},
+            "devices": devices_spec,
+        }
+        fractional = len(ids) < consts.GPU_PERCENT_EACH_CARD
+        if fractional:
+            iso = self._isolation_payload(device, "core")
+            resp["envs"].update(iso["envs"])
+            resp["mounts"] = iso["mounts"]
+        return resp
+
+    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str]):
+        n_links = self.links_for(len(ids))
+        if len(indexes) != n_links:
+            raise ValueError(
+                f"annotation has {len(indexes)} GPU indexes, expected {n_links}"
+            )
+        for i, idx in enumerate(indexes):
+            alloc_id = f"{device.hash}-{i}"
+            self.cfg.operator.create(idx, alloc_id)
+            created.append(alloc_id)
+        percent = len(ids)
+        if percent < consts.GPU_PERCENT_EACH_CARD and self.cfg.cumask and self.cfg.limits:
+            mask_hex, n_cus = self.cfg.cumask.allocate(device.hash, indexes[0], percent)
+            self.cfg.limits.finalize(
+                device.hash,
+                gpu_indexes=indexes,
+                devices=self.cfg.operator.devices(),
+                cu_mask=mask_hex,
+                cu_count=n_cus,
+            )
+        elif self.cfg.limits:
+            self.cfg.limits.finalize(
+                device.hash, gpu_indexes=indexes, devices=self.cfg.operator.devices()
+            )
+
+    def _locator(self):
+        return self.cfg.core_locator
+
+
+class GPUShareMemoryPlugin(GPUSharePluginBase):
+    """One fake device per mem_unit_mib MiB of HBM3E (288 GiB per MI355X).
+
+    The unit defaults to 1 MiB (reference contract). On 288 GB parts that is
+    294,912 device IDs per GPU; deployments that hit kubelet scaling limits
+    can set --mem-unit-mib=1024 (documented deviation, same resource name)."""
+
+    resource_name = consts.RESOURCE_GPU_MEMORY
+
+    def fake_device_ids_for_gpu(self, gpu) -> List[str]:
+        unit = self.cfg.options.mem_unit_mib
+        count = gpu.memory_mib // unit
+        return [f"{gpu.index}-{slot:06d}" for slot in range(count)]
+
+    def _allocate_one(self, device: Device, ids: List[str]) -> dict:
+        host = f"/dev/{consts.ELASTIC_GPU_LINK_FMT % (device.hash + '-0')}"
+        resp = {
+            "envs": {consts.GPU_ENV_KEY: device.hash},
+            "devices": [
+                {"container_path": consts.KFD_PATH, "host_path": consts.KFD_PATH,
+                 "permissions": "rw"},
+                {"container_path": "/dev/egpu/gpu0", "host_path": host, "permissions": "rw"},
+            ],
+        }
+        iso = self._isolation_payload(device, "mem")
+        resp["envs"].update(iso["envs"])
+        if iso["mounts"]:
+            resp["mounts"] = iso["mounts"]
+        return resp
+
+    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str]):
+        if len(indexes) != 1:
+            raise ValueError(f"memory binding expects exactly 1 GPU index, got {indexes}")
+        alloc_id = f"{device.hash}-0"
+        self.cfg.operator.create(indexes[0], alloc_id)
+        created.append(alloc_id)
+        if self.cfg.limits:
+            mem_bytes = len(device.list) * self.cfg.options.mem_unit_mib * 1024 * 1024
+            self.cfg.limits.finalize(
+                device.hash,
+                gpu_indexes=indexes,
+                devices=self.cfg.operator.devices(),
+                mem_limit_bytes=mem_bytes,
+            )
+
+    def _locator(self):
+        return self.cfg.memory_locator

@@ -36,6 +36,11 @@ class Storage:
         self._conn.execute(
             "CREATE TABLE IF NOT EXISTS pods (key TEXT PRIMARY KEY, val BLOB NOT NULL)"
         )
+        # Sidecar table for MI355X-specific metadata (CU masks, quotas) so the
+        # pods table stays byte-compatible with the reference's record format.
+        self._conn.execute(
+            "CREATE TABLE IF NOT EXISTS aux (key TEXT PRIMARY KEY, val TEXT NOT NULL)"
+        )
         self._conn.commit()
 
     @property
@@ -75,6 +80,33 @@ class Storage:
             rows = self._conn.execute("SELECT key, val FROM pods").fetchall()
         for key, val in rows:
             fn(PodInfo.from_raw(key, val))
+
+    # ---- aux KV (isolation metadata; not part of the reference format) ----
+    def aux_set(self, key: str, val: str) -> None:
+        with self._lock:
+            self._conn.execute(
+                "INSERT INTO aux(key, val) VALUES(?, ?) "
+                "ON CONFLICT(key) DO UPDATE SET val=excluded.val",
+                (key, val),
+            )
+            self._conn.commit()
+
+    def aux_get(self, key: str) -> Optional[str]:
+        with self._lock:
+            row = self._conn.execute("SELECT val FROM aux WHERE key=?", (key,)).fetchone()
+        return row[0] if row else None
+
+    def aux_delete(self, key: str) -> None:
+        with self._lock:
+            self._conn.execute("DELETE FROM aux WHERE key=?", (key,))
+            self._conn.commit()
+
+    def aux_items(self, prefix: str = "") -> list:
+        with self._lock:
+            rows = self._conn.execute(
+                "SELECT key, val FROM aux WHERE key LIKE ?", (prefix + "%",)
+            ).fetchall()
+        return rows
 
     def close(self) -> None:
         with self._lock:

@@ -1,0 +1,159 @@
+"""Test harness: an in-process agent wired to fakes + gRPC client helpers."""
+from __future__ import annotations
+
+import os
+import threading
+from concurrent import futures
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import grpc
+
+from elastic_gpu_agent_amd import consts
+from elastic_gpu_agent_amd.isolation import CUMaskAllocator, LimitsWriter
+from elastic_gpu_agent_amd.kube.locator import FakeDeviceLocator
+from elastic_gpu_agent_amd.kube.pods import Pod
+from elastic_gpu_agent_amd.kube.sitter import FakeSitter
+from elastic_gpu_agent_amd.operator import GPUOperator
+from elastic_gpu_agent_amd.operator.fake import FakeBackend
+from elastic_gpu_agent_amd.plugins.aggregate import GPUSharePlugin
+from elastic_gpu_agent_amd.plugins.config import AgentPaths, GPUPluginConfig, PluginOptions
+from elastic_gpu_agent_amd.protos import deviceplugin as dp
+from elastic_gpu_agent_amd.storage import Storage
+
+
+@dataclass
+class Harness:
+    tmp: str
+    gpus: int = 2
+    mem_unit_mib: int = 1024  # keep fake-device counts small in tests
+    storage: Storage = None
+    operator: GPUOperator = None
+    sitter: FakeSitter = None
+    core_locator: FakeDeviceLocator = None
+    mem_locator: FakeDeviceLocator = None
+    plugin: GPUSharePlugin = None
+    paths: AgentPaths = None
+
+    def __post_init__(self):
+        dev_root = os.path.join(self.tmp, "dev")
+        limits_dir = os.path.join(self.tmp, "limits")
+        plugin_dir = os.path.join(self.tmp, "device-plugins")
+        os.makedirs(plugin_dir, exist_ok=True)
+        self.storage = Storage(os.path.join(self.tmp, "meta.db"))
+        self.operator = GPUOperator(FakeBackend(count=self.gpus), dev_root=dev_root)
+        self.sitter = FakeSitter()
+        self.core_locator = FakeDeviceLocator()
+        self.mem_locator = FakeDeviceLocator()
+        self.paths = AgentPaths(
+            dev_root=dev_root,
+            plugin_dir=plugin_dir,
+            kubelet_socket=os.path.join(plugin_dir, "kubelet.sock"),
+            limits_dir=limits_dir,
+            shim_host_path=os.path.join(self.tmp, "libegpu_shim.so"),
+        )
+        cfg = GPUPluginConfig(
+            operator=self.operator,
+            storage=self.storage,
+            sitter=self.sitter,
+            core_locator=self.core_locator,
+            memory_locator=self.mem_locator,
+            paths=self.paths,
+            options=PluginOptions(mem_unit_mib=self.mem_unit_mib),
+            limits=LimitsWriter(limits_dir),
+            cumask=CUMaskAllocator(self.storage, self.operator.devices()),
+        )
+        self.plugin = GPUSharePlugin(cfg)
+
+    def close(self):
+        self.plugin.stop()
+        self.storage.close()
+
+    # -- scenario helpers --
+    def add_assumed_pod(self, ns, name, container, gpu_indexes: str):
+        pod = Pod(
+            namespace=ns,
+            name=name,
+            annotations={
+                consts.ELASTIC_GPU_ASSUMED_ANNOTATION: "true",
+                consts.ELASTIC_GPU_CONTAINER_ANNOTATION % container: gpu_indexes,
+            },
+        )
+        self.sitter.add(pod)
+        return pod
+
+
+class PluginClient:
+    """gRPC client speaking the device-plugin API to a served socket."""
+
+    def __init__(self, socket_path: str):
+        self.channel = grpc.insecure_channel(f"unix://{socket_path}")
+        mk = self.channel.unary_unary
+        self.get_options = mk(
+            dp.METHOD_GET_OPTIONS,
+            request_serializer=dp.Empty.encode,
+            response_deserializer=dp.DevicePluginOptions.decode,
+        )
+        self.allocate = mk(
+            dp.METHOD_ALLOCATE,
+            request_serializer=dp.AllocateRequest.encode,
+            response_deserializer=dp.AllocateResponse.decode,
+        )
+        self.pre_start = mk(
+            dp.METHOD_PRE_START_CONTAINER,
+            request_serializer=dp.PreStartContainerRequest.encode,
+            response_deserializer=dp.PreStartContainerResponse.decode,
+        )
+        self.preferred = mk(
+            dp.METHOD_GET_PREFERRED_ALLOCATION,
+            request_serializer=dp.PreferredAllocationRequest.encode,
+            response_deserializer=dp.PreferredAllocationResponse.decode,
+        )
+        self.list_and_watch = self.channel.unary_stream(
+            dp.METHOD_LIST_AND_WATCH,
+            request_serializer=dp.Empty.encode,
+            response_deserializer=dp.ListAndWatchResponse.decode,
+        )
+
+    def close(self):
+        self.channel.close()
+
+
+class FakeKubeletRegistration:
+    """Records device-plugin Register calls (stand-in kubelet.sock server)."""
+
+    def __init__(self, socket_path: str):
+        self.requests: List[dict] = []
+        self._event = threading.Event()
+        handler = grpc.method_handlers_generic_handler(
+            dp.REGISTRATION_SERVICE,
+            {
+                "Register": grpc.unary_unary_rpc_method_handler(
+                    self._register,
+                    request_deserializer=dp.RegisterRequest.decode,
+                    response_serializer=dp.Empty.encode,
+                )
+            },
+        )
+        self.server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+        self.server.add_generic_rpc_handlers((handler,))
+        self.server.add_insecure_port(f"unix://{socket_path}")
+
+    def _register(self, request, context):
+        self.requests.append(request)
+        self._event.set()
+        return {}
+
+    def wait_for_register(self, n=1, timeout=10.0) -> bool:
+        import time
+
+        deadline = time.time() + timeout
+        while len(self.requests) < n and time.time() < deadline:
+            time.sleep(0.05)
+        return len(self.requests) >= n
+
+    def start(self):
+        self.server.start()
+
+    def stop(self):
+        self.server.stop(grace=0.2)

@@ -1,0 +1,224 @@
+"""Device-plugin handler tests (in-process, fake backend): advertisement,
+Allocate, PreStart binding, isolation wiring, GC, Restore."""
+import json
+import os
+
+import pytest
+
+from elastic_gpu_agent_amd import consts
+from elastic_gpu_agent_amd.types import Device
+
+from helpers import Harness
+
+
+@pytest.fixture
+def h(tmp_path):
+    harness = Harness(str(tmp_path), gpus=2)
+    yield harness
+    harness.close()
+
+
+def test_core_advertises_100_per_gpu_with_topology(h):
+    devices = h.plugin.core.list_devices()
+    assert len(devices) == 200
+    assert devices[0]["ID"] == "0-00" and devices[199]["ID"] == "1-99"
+    assert all(d["health"] == "Healthy" for d in devices)
+    assert devices[0]["topology"]["nodes"][0]["ID"] == 0
+
+
+def test_memory_advertises_units(h):
+    devices = h.plugin.memory.list_devices()
+    # 288 GiB per GPU at 1024 MiB units = 288 per GPU
+    assert len(devices) == 2 * 288
+    assert devices[0]["ID"] == "0-000000"
+
+
+def test_options(h):
+    opts = h.plugin.core.get_device_plugin_options({}, None)
+    assert opts["pre_start_required"] is True
+    assert opts["get_preferred_allocation_available"] is True
+
+
+def test_core_allocate_fractional(h):
+    ids = [f"0-{i:02d}" for i in range(30)]
+    resp = h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    assert len(resp["container_responses"]) == 1
+    cr = resp["container_responses"][0]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    assert cr["envs"][consts.GPU_ENV_KEY] == d.hash
+    # fractional => shim injected
+    assert cr["envs"]["HSA_TOOLS_LIB"] == h.paths.shim_container_path
+    assert any(m["container_path"] == "/etc/egpu/limits-core.json" for m in cr["mounts"])
+    # kfd + one per-alloc device link
+    paths = [s["host_path"] for s in cr["devices"]]
+    assert consts.KFD_PATH in paths
+    assert f"/dev/elastic-gpu-{d.hash}-0" in paths
+    # limits file exists (touched) so kubelet can mount it
+    assert os.path.exists(h.plugin.cfg.limits.host_path(d.hash))
+
+
+def test_core_allocate_whole_gpus_no_shim(h):
+    ids = [f"0-{i:02d}" for i in range(100)] + [f"1-{i:02d}" for i in range(100)]
+    resp = h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    cr = resp["container_responses"][0]
+    assert "HSA_TOOLS_LIB" not in cr["envs"]
+    # 2 GPU links + kfd
+    assert len(cr["devices"]) == 3
+
+
+def test_core_allocate_per_container_responses(h):
+    """Multi-container pods get one response per container (reference merged
+    them into one — SURVEY §3.3; deliberately not copied)."""
+    r1 = [f"0-{i:02d}" for i in range(10)]
+    r2 = [f"1-{i:02d}" for i in range(20)]
+    resp = h.plugin.core.allocate(
+        {"container_requests": [{"devicesIDs": r1}, {"devicesIDs": r2}]}, None
+    )
+    assert len(resp["container_responses"]) == 2
+    h1 = Device.new(r1, consts.RESOURCE_GPU_CORE).hash
+    h2 = Device.new(r2, consts.RESOURCE_GPU_CORE).hash
+    assert resp["container_responses"][0]["envs"]["GPU"] == h1
+    assert resp["container_responses"][1]["envs"]["GPU"] == h2
+
+
+def test_core_prestart_binds(h):
+    ids = [f"0-{i:02d}" for i in range(25)]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    h.core_locator.assign(d.hash, PodContainer("ns", "p1", "main"))
+    h.add_assumed_pod("ns", "p1", "main", "0")
+    h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+
+    # symlinks materialized: gpu link -> renderD128 (GPU 0), ctl -> kfd
+    gpu_link = os.path.join(h.paths.dev_root, f"elastic-gpu-{d.hash}-0")
+    ctl_link = os.path.join(h.paths.dev_root, f"elastic-gpuctl-{d.hash}-0")
+    assert os.readlink(gpu_link) == "/dev/dri/renderD128"
+    assert os.readlink(ctl_link) == "/dev/kfd"
+
+    # state persisted in the reference's record format
+    pi = h.storage.load("ns", "p1")
+    assert pi.container_device_map["main"].hash == d.hash
+
+    # CU mask allocated: 25% -> 64 CUs spread over XCDs, limits finalized
+    rec = json.loads(h.storage.aux_get("mask/" + d.hash))
+    assert rec["cu_count"] == 64 and rec["gpu_index"] == 0
+    limits = h.plugin.cfg.limits.read(d.hash)
+    assert limits["cu_count"] == 64
+    assert limits["render_minors"] == [128]
+    assert limits["cu_mask"] == rec["cu_mask"]
+
+
+def test_prestart_rejects_unassumed(h):
+    ids = [f"0-{i:02d}" for i in range(10)]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    from elastic_gpu_agent_amd.kube.pods import Pod
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    h.core_locator.assign(d.hash, PodContainer("ns", "p2", "main"))
+    h.sitter.add(Pod(namespace="ns", name="p2", annotations={}))  # not assumed
+    with pytest.raises(RuntimeError, match="assumed"):
+        h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+
+
+def test_prestart_rollback_on_bad_annotation(h):
+    ids = [f"0-{i:02d}" for i in range(150)]  # needs 2 GPU indexes
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    h.core_locator.assign(d.hash, PodContainer("ns", "p3", "main"))
+    h.add_assumed_pod("ns", "p3", "main", "0")  # only 1 index -> mismatch
+    with pytest.raises(RuntimeError, match="bind"):
+        h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+    # no stray symlinks survive the rollback
+    assert not os.path.exists(os.path.join(h.paths.dev_root, f"elastic-gpu-{d.hash}-0"))
+
+
+def test_memory_prestart_sets_quota(h):
+    ids = [f"1-{i:06d}" for i in range(72)]  # 72 GiB at 1024 MiB units
+    d = Device.new(ids, consts.RESOURCE_GPU_MEMORY)
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    h.mem_locator.assign(d.hash, PodContainer("ns", "pm", "main"))
+    h.add_assumed_pod("ns", "pm", "main", "1")
+    h.plugin.memory.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    h.plugin.memory.pre_start_container({"devicesIDs": ids}, None)
+    limits = h.plugin.cfg.limits.read(d.hash)
+    assert limits["mem_limit_bytes"] == 72 * 1024**3
+    assert limits["render_minors"] == [129]  # GPU 1
+    gpu_link = os.path.join(h.paths.dev_root, f"elastic-gpu-{d.hash}-0")
+    assert os.readlink(gpu_link) == "/dev/dri/renderD129"
+
+
+def test_gc_reclaims_deleted_pod(h):
+    ids = [f"0-{i:02d}" for i in range(25)]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    h.core_locator.assign(d.hash, PodContainer("ns", "pgc", "main"))
+    h.add_assumed_pod("ns", "pgc", "main", "0")
+    h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+
+    assert h.plugin.gc_once() == 0  # pod alive -> nothing reclaimed
+
+    h.sitter.remove("ns", "pgc")
+    assert h.plugin.gc_once() == 1
+    assert not os.path.exists(os.path.join(h.paths.dev_root, f"elastic-gpu-{d.hash}-0"))
+    assert h.storage.aux_get("mask/" + d.hash) is None
+    assert not os.path.exists(h.plugin.cfg.limits.host_path(d.hash))
+    with pytest.raises(KeyError):
+        h.storage.load("ns", "pgc")
+
+
+def test_restore_recreates_links(h):
+    ids = [f"0-{i:02d}" for i in range(25)]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    h.core_locator.assign(d.hash, PodContainer("ns", "pr", "main"))
+    h.add_assumed_pod("ns", "pr", "main", "0")
+    h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+
+    # simulate node reboot: /host/dev wiped
+    gpu_link = os.path.join(h.paths.dev_root, f"elastic-gpu-{d.hash}-0")
+    os.unlink(gpu_link)
+    os.unlink(os.path.join(h.paths.dev_root, f"elastic-gpuctl-{d.hash}-0"))
+
+    restored = h.plugin.restore()
+    assert restored == 1
+    assert os.readlink(gpu_link) == "/dev/dri/renderD128"
+
+
+def test_preferred_allocation_handler(h):
+    avail = [f"0-{i:02d}" for i in range(30)] + [f"1-{i:02d}" for i in range(100)]
+    resp = h.plugin.core.get_preferred_allocation(
+        {
+            "container_requests": [
+                {"available_deviceIDs": avail, "must_include_deviceIDs": [], "allocation_size": 10}
+            ]
+        },
+        None,
+    )
+    picked = resp["container_responses"][0]["deviceIDs"]
+    assert len(picked) == 10 and all(p.startswith("0-") for p in picked)
+
+
+def test_cumask_disjoint_across_pods(h):
+    """Two fractional pods on the same GPU get disjoint CU masks."""
+    from elastic_gpu_agent_amd.isolation.cumask import parse_mask_hex
+    from elastic_gpu_agent_amd.types import PodContainer
+
+    masks = []
+    for i, pct in enumerate((25, 50)):
+        ids = [f"0-{k:02d}" for k in range(pct)]
+        d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+        h.core_locator.assign(d.hash, PodContainer("ns", f"pd{i}", "main"))
+        h.add_assumed_pod("ns", f"pd{i}", "main", "0")
+        h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+        h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+        masks.append(parse_mask_hex(h.plugin.cfg.limits.read(d.hash)["cu_mask"]))
+    a, b = masks
+    assert all((wa & wb) == 0 for wa, wb in zip(a, b))

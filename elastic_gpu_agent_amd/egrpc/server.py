@@ -1,0 +1,490 @@
+"""egrpc server: a purpose-built gRPC-over-HTTP/2 server on unix sockets.
+
+Why it exists: grpcio's Python server costs ~1.2 ms per unary RPC on this
+class of host (thread-pool handoffs), two orders of magnitude above the raw
+socket round trip (~40 µs). The kubelet device-plugin surface is tiny — four
+unary methods + one server-stream per resource — and Allocate p50 latency is
+the headline metric, so the transport is implemented directly:
+
+- one thread per connection; **unary handlers dispatch inline** on the
+  connection thread (no queue, no pool) — response headers and OK-trailers
+  are precomputed HPACK blocks, so the hot path is: parse frames → decode
+  :path → handler → one sendall;
+- server-streaming handlers (ListAndWatch blocks for minutes) run on their
+  own thread per stream, writes serialized by a per-connection lock;
+- full HPACK (dynamic table + Huffman) on the receive side for interop with
+  Go kubelet / grpcio clients (validated in tests against grpcio);
+- flow control: peers get a large connection/stream window up front
+  (RECV_WINDOW) and replenishment after every DATA frame; outbound DATA
+  respects the peer's windows, pumping incoming frames while blocked.
+
+Interop is covered by tests/test_egrpc.py (grpcio client ↔ this server and
+the reverse) including >1 MiB messages and streaming.
+"""
+from __future__ import annotations
+
+import logging
+import os
+import socket
+import struct
+import threading
+from typing import Callable, Dict, List, Optional, Tuple
+
+from . import core, hpack
+from .core import (
+    ConnectionClosed,
+    EgrpcError,
+    frame_header,
+    parse_frame_header,
+)
+
+log = logging.getLogger(__name__)
+
+# precomputed response blocks (stateless encoder → constant bytes)
+_RESP_HEADERS_BLOCK = hpack.encode_headers(
+    [(b":status", b"200"), (b"content-type", b"application/grpc")]
+)
+_RESP_HEADERS = (
+    frame_header(len(_RESP_HEADERS_BLOCK), core.HEADERS, core.FLAG_END_HEADERS, 0)[:5]
+)  # stream id patched per use
+_OK_TRAILERS_BLOCK = hpack.encode_headers([(b"grpc-status", b"0")])
+
+
+def _headers_frame(block: bytes, stream_id: int, flags: int) -> bytes:
+    return frame_header(len(block), core.HEADERS, flags, stream_id) + block
+
+
+class Method:
+    __slots__ = ("fn", "request_deserializer", "response_serializer", "server_streaming")
+
+    def __init__(self, fn, request_deserializer, response_serializer, server_streaming=False):
+        self.fn = fn
+        self.request_deserializer = request_deserializer
+        self.response_serializer = response_serializer
+        self.server_streaming = server_streaming
+
+
+def unary_unary(fn, request_deserializer=None, response_serializer=None) -> Method:
+    return Method(fn, request_deserializer, response_serializer, False)
+
+
+def unary_stream(fn, request_deserializer=None, response_serializer=None) -> Method:
+    return Method(fn, request_deserializer, response_serializer, True)
+
+
+class ServerContext:
+    """Handler-visible context (subset of grpc.ServicerContext the agent uses)."""
+
+    def __init__(self, conn: "_Connection", stream_id: int):
+        self._conn = conn
+        self._stream_id = stream_id
+        self.cancelled = threading.Event()
+
+    def is_active(self) -> bool:
+        return not self.cancelled.is_set() and not self._conn.closed
+
+    def abort(self, code, message: str = ""):
+        code_int = getattr(code, "value", code)
+        if isinstance(code_int, tuple):  # grpc.StatusCode enum value: (int, str)
+            code_int = code_int[0]
+        raise EgrpcError(int(code_int), message)
+
+
+class _Stream:
+    __slots__ = ("id", "path", "data", "end_stream", "headers", "context", "data_hdr")
+
+    def __init__(self, sid: int):
+        self.id = sid
+        self.path: bytes = b""
+        self.data = bytearray()
+        self.end_stream = False
+        self.headers: List[Tuple[bytes, bytes]] = []
+        self.context: Optional[ServerContext] = None
+
+
+class _Connection:
+    def __init__(self, server: "Server", sock: socket.socket):
+        self.server = server
+        self.sock = sock
+        self.closed = False
+        self.buf = bytearray()
+        self.decoder = hpack.Decoder()
+        self.streams: Dict[int, _Stream] = {}
+        self.write_lock = threading.Lock()
+        # flow control
+        self.peer_max_frame = core.DEFAULT_MAX_FRAME
+        self.peer_initial_window = core.DEFAULT_WINDOW
+        self.conn_send_window = core.DEFAULT_WINDOW
+        self.stream_send_windows: Dict[int, int] = {}
+        self.window_cv = threading.Condition()
+        self._hdr_accum: Optional[_Stream] = None  # awaiting CONTINUATION
+        self.conn_recv_deficit = 0
+        self.stream_recv_deficit: Dict[int, int] = {}
+
+    # ---- writing ----
+    def send(self, data: bytes) -> None:
+        with self.write_lock:
+            try:
+                self.sock.sendall(data)
+            except OSError:
+                self.closed = True
+
+    def send_data_frames(self, stream_id: int, payload: bytes, end_stream: bool,
+                         pump: Optional[Callable[[], None]] = None) -> None:
+        """Send DATA respecting peer windows. `pump` processes incoming frames
+        while blocked on flow control (connection thread passes its reader)."""
+        view = memoryview(payload)
+        off, total = 0, len(payload)
+        while off < total or (total == 0 and end_stream):
+            avail = min(self.conn_send_window,
+                        self.stream_send_windows.get(stream_id, self.peer_initial_window))
+            if total > 0 and avail <= 0:
+                if pump is not None:
+                    pump()  # reader thread: process one incoming frame inline
+                else:
+                    with self.window_cv:
+                        self.window_cv.wait(timeout=0.1)
+                if self.closed:
+                    raise ConnectionClosed()
+                continue
+            n = min(total - off, avail if total else 0, self.peer_max_frame)
+            if total == 0:
+                n = 0
+            last = off + n >= total
+            flags = core.FLAG_END_STREAM if (end_stream and last) else 0
+            chunk = bytes(view[off : off + n])
+            self.send(frame_header(n, core.DATA, flags, stream_id) + chunk)
+            self.conn_send_window -= n
+            self.stream_send_windows[stream_id] = (
+                self.stream_send_windows.get(stream_id, self.peer_initial_window) - n
+            )
+            off += n
+            if last:
+                break
+
+    # ---- unary fast path: headers+data+trailers in one write ----
+    def send_unary_response(self, stream_id: int, message: bytes) -> None:
+        payload = core.grpc_frame(message)
+        fits = (len(payload) <= self.conn_send_window
+                and len(payload) <= self.stream_send_windows.get(
+                    stream_id, self.peer_initial_window)
+                and len(payload) <= self.peer_max_frame)
+        headers = _headers_frame(_RESP_HEADERS_BLOCK, stream_id, core.FLAG_END_HEADERS)
+        trailers = _headers_frame(
+            _OK_TRAILERS_BLOCK, stream_id, core.FLAG_END_HEADERS | core.FLAG_END_STREAM
+        )
+        if fits:
+            data = frame_header(len(payload), core.DATA, 0, stream_id) + payload
+            self.send(headers + data + trailers)
+            self.conn_send_window -= len(payload)
+            self.stream_send_windows[stream_id] = (
+                self.stream_send_windows.get(stream_id, self.peer_initial_window)
+                - len(payload)
+            )
+        else:
+            self.send(headers)
+            self.send_data_frames(stream_id, payload, end_stream=False,
+                                  pump=self._pump_one_frame)
+            self.send(trailers)
+
+    def send_error(self, stream_id: int, code: int, message: str) -> None:
+        block = hpack.encode_headers(
+            [(b":status", b"200"), (b"content-type", b"application/grpc"),
+             (b"grpc-status", str(code).encode()),
+             (b"grpc-message", core.percent_encode(message))]
+        )
+        self.send(_headers_frame(block, stream_id,
+                                 core.FLAG_END_HEADERS | core.FLAG_END_STREAM))
+
+    # ---- reading ----
+    def _read_frame(self):
+        hdr = core.read_exact(self.sock, 9, self.buf)
+        length, ftype, flags, sid = parse_frame_header(hdr)
+        body = core.read_exact(self.sock, length, self.buf) if length else b""
+        return ftype, flags, sid, body
+
+    def _pump_one_frame(self) -> None:
+        ftype, flags, sid, body = self._read_frame()
+        self._process_frame(ftype, flags, sid, body, defer_dispatch=True)
+
+    _deferred: List[_Stream]
+
+    def run(self) -> None:
+        self._deferred = []
+        try:
+            core.read_exact(self.sock, len(core.PREFACE), self.buf)
+            # our SETTINGS + a big connection receive window
+            self.send(
+                core.settings_frame([
+                    (core.SETTINGS_MAX_FRAME_SIZE, core.OUR_MAX_FRAME),
+                    (core.SETTINGS_INITIAL_WINDOW_SIZE, core.RECV_WINDOW),
+                    (core.SETTINGS_MAX_CONCURRENT_STREAMS, 1024),
+                ])
+                + core.window_update(0, core.RECV_WINDOW - core.DEFAULT_WINDOW)
+            )
+            while not self.closed:
+                ftype, flags, sid, body = self._read_frame()
+                self._process_frame(ftype, flags, sid, body, defer_dispatch=False)
+                while self._deferred:
+                    st = self._deferred.pop(0)
+                    self._dispatch(st)
+        except (ConnectionClosed, OSError):
+            pass
+        except Exception as e:  # defensive: a protocol bug must not kill the server
+            log.error("egrpc connection error: %s", e, exc_info=True)
+        finally:
+            self.closed = True
+            for st in self.streams.values():
+                if st.context is not None:
+                    st.context.cancelled.set()
+            try:
+                self.sock.close()
+            except OSError:
+                pass
+
+    def _process_frame(self, ftype, flags, sid, body, defer_dispatch: bool) -> None:
+        if ftype == core.SETTINGS:
+            if not flags & core.FLAG_ACK:
+                for off in range(0, len(body) - 5, 6):
+                    k, v = struct.unpack_from(">HI", body, off)
+                    if k == core.SETTINGS_MAX_FRAME_SIZE:
+                        self.peer_max_frame = v
+                    elif k == core.SETTINGS_INITIAL_WINDOW_SIZE:
+                        delta = v - self.peer_initial_window
+                        self.peer_initial_window = v
+                        for s in self.stream_send_windows:
+                            self.stream_send_windows[s] += delta
+                self.send(core.settings_frame([], flags=core.FLAG_ACK))
+        elif ftype == core.PING:
+            if not flags & core.FLAG_ACK:
+                self.send(frame_header(8, core.PING, core.FLAG_ACK, 0) + body)
+        elif ftype == core.WINDOW_UPDATE:
+            (inc,) = struct.unpack(">I", body)
+            if sid == 0:
+                self.conn_send_window += inc
+            else:
+                self.stream_send_windows[sid] = (
+                    self.stream_send_windows.get(sid, self.peer_initial_window) + inc
+                )
+            with self.window_cv:
+                self.window_cv.notify_all()
+        elif ftype in (core.HEADERS, core.CONTINUATION):
+            if ftype == core.HEADERS:
+                st = self.streams.get(sid)
+                if st is None:
+                    st = self.streams[sid] = _Stream(sid)
+                pad = 0
+                off = 0
+                if flags & core.FLAG_PADDED:
+                    pad = body[0]
+                    off = 1
+                if flags & core.FLAG_PRIORITY:
+                    off += 5
+                block = body[off : len(body) - pad]
+                st.end_stream = st.end_stream or bool(flags & core.FLAG_END_STREAM)
+                st.headers += self.decoder.decode(block) if flags & core.FLAG_END_HEADERS \
+                    else []
+                if not flags & core.FLAG_END_HEADERS:
+                    # accumulate into CONTINUATION
+                    st.data_hdr = bytearray(block)  # type: ignore[attr-defined]
+                    self._hdr_accum = st
+                    return
+            else:  # CONTINUATION
+                st = self._hdr_accum
+                if st is None:
+                    return
+                st.data_hdr += body  # type: ignore[attr-defined]
+                if not flags & core.FLAG_END_HEADERS:
+                    return
+                st.headers += self.decoder.decode(bytes(st.data_hdr))  # type: ignore
+                self._hdr_accum = None
+            for name, value in st.headers:
+                if name == b":path":
+                    st.path = value
+            if st.end_stream:
+                self._queue_dispatch(st, defer_dispatch)
+        elif ftype == core.DATA:
+            st = self.streams.get(sid)
+            if st is None:
+                return
+            pad = 0
+            off = 0
+            if flags & core.FLAG_PADDED:
+                pad = body[0]
+                off = 1
+            st.data += body[off : len(body) - pad]
+            if len(body):
+                # replenish lazily: our advertised windows are RECV_WINDOW;
+                # batched updates once half is consumed save two syscalls per
+                # small request (streams are short-lived except ListAndWatch,
+                # which receives almost nothing)
+                self.conn_recv_deficit += len(body)
+                upd = b""
+                if self.conn_recv_deficit >= core.RECV_WINDOW // 2:
+                    upd += core.window_update(0, self.conn_recv_deficit)
+                    self.conn_recv_deficit = 0
+                sd = self.stream_recv_deficit.get(sid, 0) + len(body)
+                if sd >= core.RECV_WINDOW // 2 and not st.end_stream:
+                    upd += core.window_update(sid, sd)
+                    sd = 0
+                self.stream_recv_deficit[sid] = sd
+                if upd:
+                    self.send(upd)
+            if flags & core.FLAG_END_STREAM:
+                st.end_stream = True
+                self._queue_dispatch(st, defer_dispatch)
+        elif ftype == core.RST_STREAM:
+            st = self.streams.get(sid)
+            if st is not None and st.context is not None:
+                st.context.cancelled.set()
+            self._close_stream(sid)
+        elif ftype == core.GOAWAY:
+            self.closed = True
+        # PRIORITY / PUSH_PROMISE / unknown: ignore
+
+    def _close_stream(self, sid: int) -> None:
+        self.streams.pop(sid, None)
+        self.stream_recv_deficit.pop(sid, None)
+
+    def _queue_dispatch(self, st: _Stream, defer: bool) -> None:
+        if defer:
+            self._deferred.append(st)
+        else:
+            self._dispatch(st)
+
+    def _dispatch(self, st: _Stream) -> None:
+        method = self.server.methods.get(st.path.decode())
+        if method is None:
+            self.send_error(st.id, core.UNKNOWN, f"unknown method {st.path.decode()!r}")
+            self._close_stream(st.id)
+            return
+        try:
+            msgs = core.parse_grpc_frames(bytes(st.data))
+            raw = msgs[0] if msgs else b""
+            request = method.request_deserializer(raw) if method.request_deserializer else raw
+        except Exception as e:
+            self.send_error(st.id, core.INTERNAL, f"bad request: {e}")
+            self._close_stream(st.id)
+            return
+        ctx = ServerContext(self, st.id)
+        st.context = ctx
+        if method.server_streaming:
+            t = threading.Thread(
+                target=self._run_streaming, args=(method, request, ctx, st),
+                name=f"egrpc-stream-{st.id}", daemon=True,
+            )
+            t.start()
+            return
+        # unary inline fast path
+        try:
+            resp = method.fn(request, ctx)
+            payload = method.response_serializer(resp) if method.response_serializer else resp
+        except EgrpcError as e:
+            self.send_error(st.id, e.code(), e.details())
+            self._close_stream(st.id)
+            return
+        except Exception as e:
+            log.error("handler error on %s: %s", st.path, e, exc_info=True)
+            self.send_error(st.id, core.UNKNOWN, str(e))
+            self._close_stream(st.id)
+            return
+        self.send_unary_response(st.id, payload)
+        self._close_stream(st.id)
+
+    def _run_streaming(self, method: Method, request, ctx: ServerContext, st: _Stream) -> None:
+        sid = st.id
+        try:
+            self.send(_headers_frame(_RESP_HEADERS_BLOCK, sid, core.FLAG_END_HEADERS))
+            for item in method.fn(request, ctx):
+                if not ctx.is_active():
+                    break
+                payload = method.response_serializer(item) if method.response_serializer \
+                    else item
+                self.send_data_frames(sid, core.grpc_frame(payload), end_stream=False)
+            if not self.closed:
+                self.send(_headers_frame(
+                    _OK_TRAILERS_BLOCK, sid,
+                    core.FLAG_END_HEADERS | core.FLAG_END_STREAM))
+        except EgrpcError as e:
+            if not self.closed:
+                block = hpack.encode_headers(
+                    [(b"grpc-status", str(e.code()).encode()),
+                     (b"grpc-message", core.percent_encode(e.details()))])
+                self.send(_headers_frame(
+                    block, sid, core.FLAG_END_HEADERS | core.FLAG_END_STREAM))
+        except (ConnectionClosed, OSError):
+            pass
+        except Exception as e:
+            log.error("streaming handler error on %s: %s", st.path, e, exc_info=True)
+            if not self.closed:
+                self.send_error(sid, core.UNKNOWN, str(e))
+        finally:
+            self._close_stream(sid)
+
+
+class Server:
+    """gRPC server over a unix socket. API shape: add methods, start, stop."""
+
+    def __init__(self):
+        self.methods: Dict[str, Method] = {}
+        self._sock: Optional[socket.socket] = None
+        self._path: Optional[str] = None
+        self._accept_thread: Optional[threading.Thread] = None
+        self._conns: List[_Connection] = []
+        self._stopped = threading.Event()
+
+    def add_method(self, path: str, method: Method) -> None:
+        self.methods[path] = method
+
+    def add_service(self, service_name: str, methods: Dict[str, Method]) -> None:
+        for name, m in methods.items():
+            self.methods[f"/{service_name}/{name}"] = m
+
+    def bind_unix(self, path: str) -> None:
+        if os.path.exists(path):
+            os.unlink(path)
+        sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        sock.bind(path)
+        sock.listen(128)
+        self._sock = sock
+        self._path = path
+
+    def start(self) -> None:
+        assert self._sock is not None, "bind_unix first"
+        self._accept_thread = threading.Thread(
+            target=self._accept_loop, name="egrpc-accept", daemon=True
+        )
+        self._accept_thread.start()
+
+    def _accept_loop(self) -> None:
+        while not self._stopped.is_set():
+            try:
+                conn_sock, _ = self._sock.accept()
+            except OSError:
+                return
+            conn = _Connection(self, conn_sock)
+            self._conns.append(conn)
+            threading.Thread(target=conn.run, name="egrpc-conn", daemon=True).start()
+
+    def stop(self, grace: float = 0.0) -> None:
+        self._stopped.set()
+        if self._sock is not None:
+            try:
+                self._sock.close()
+            except OSError:
+                pass
+        for conn in self._conns:
+            conn.closed = True
+            try:
+                conn.sock.shutdown(socket.SHUT_RDWR)
+            except OSError:
+                pass
+            try:
+                conn.sock.close()
+            except OSError:
+                pass
+        if self._path and os.path.exists(self._path):
+            try:
+                os.unlink(self._path)
+            except OSError:
+                pass

@@ -65,35 +65,47 @@ class CUMaskAllocator:
         from .cumask import cu_count_for_percent
 
         n = cu_count_for_percent(percent, total)
+        pairs_per_xcd = per_xcd // 2
         with self._lock:
             used = self._live_cus(gpu_index)
-            base, extra = divmod(n, xcds)
+            # ROCr CU masks have pair granularity: allocate whole CU pairs.
+            base, extra = divmod((n + 1) // 2, xcds)
             cus: List[int] = []
             for xcd in range(xcds):
                 want = base + (1 if xcd < extra else 0)
-                free = [
-                    xcd * per_xcd + k for k in range(per_xcd) if xcd * per_xcd + k not in used
+                free_pairs = [
+                    xcd * per_xcd + 2 * p
+                    for p in range(pairs_per_xcd)
+                    if xcd * per_xcd + 2 * p not in used
+                    and xcd * per_xcd + 2 * p + 1 not in used
                 ]
-                take = free[:want]
+                take = free_pairs[:want]
                 if len(take) < want:
-                    # oversubscribed: wrap around, overlapping least-recently
-                    # chosen CUs of this XCD
-                    all_cus = [xcd * per_xcd + k for k in range(per_xcd)]
-                    for c in all_cus:
+                    # oversubscribed: wrap around, overlapping already-used
+                    # pairs of this XCD
+                    for p in range(pairs_per_xcd):
                         if len(take) >= want:
                             break
-                        if c not in take:
-                            take.append(c)
-                cus.extend(take)
+                        cu0 = xcd * per_xcd + 2 * p
+                        if cu0 not in take:
+                            take.append(cu0)
+                for cu0 in take:
+                    cus.extend((cu0, cu0 + 1))
+            n_eff = len(cus)  # pair rounding may add one CU over the ask
             words = mask_words_from_cus(cus, total)
             hexmask = mask_hex(words)
             self._storage.aux_set(
                 AUX_MASK_PREFIX + alloc_hash,
                 json.dumps(
-                    {"gpu_index": gpu_index, "cu_mask": hexmask, "cu_count": n, "percent": percent}
+                    {
+                        "gpu_index": gpu_index,
+                        "cu_mask": hexmask,
+                        "cu_count": n_eff,
+                        "percent": percent,
+                    }
                 ),
             )
-        return hexmask, n
+        return hexmask, n_eff
 
     def release(self, alloc_hash: str) -> None:
         self._storage.aux_delete(AUX_MASK_PREFIX + alloc_hash)

@@ -8,7 +8,10 @@ XCDs** so an X% pod gets ≈X% of every XCD's CUs (and therefore ≈X% of
 aggregate L2 and memory bandwidth).
 
 The mask format matches ``hsa_amd_queue_cu_set_mask`` (hsa_ext_amd.h:1359):
-little-endian array of uint32, bit i = CU i enabled.
+little-endian array of uint32, bit i = CU i enabled. ROCr requires masks at
+**CU-pair granularity** (bits 2k and 2k+1 set together — the header's
+"0x33 valid, 0x5/0x6 invalid" rule), so allocation works in pairs: 128 pairs
+per GPU, 16 per XCD.
 """
 from __future__ import annotations
 
@@ -18,12 +21,13 @@ from .. import consts
 
 
 def cu_count_for_percent(percent: int, total_cus: int = consts.GFX950_CU_COUNT) -> int:
-    """gpu-core units (percent of a card) → CU count, minimum 1, rounded to
-    nearest CU."""
+    """gpu-core units (percent of a card) → CU count, minimum one CU pair,
+    rounded UP to even (ROCr masks have CU-pair granularity)."""
     if percent >= consts.GPU_PERCENT_EACH_CARD:
         return total_cus
-    return max(1, (percent * total_cus + consts.GPU_PERCENT_EACH_CARD // 2)
-               // consts.GPU_PERCENT_EACH_CARD)
+    n = max(1, (percent * total_cus + consts.GPU_PERCENT_EACH_CARD // 2)
+            // consts.GPU_PERCENT_EACH_CARD)
+    return min(total_cus, n + (n & 1))
 
 
 def xcd_round_robin_cus(
@@ -32,19 +36,25 @@ def xcd_round_robin_cus(
     xcd_count: int = consts.GFX950_XCD_COUNT,
     offset: int = 0,
 ) -> List[int]:
-    """Pick ``n_cus`` physical CU ids spread evenly across XCDs.
+    """Pick ``n_cus`` (rounded up to even) physical CU ids spread evenly
+    across XCDs, in whole CU pairs.
 
     CU ids are laid out XCD-major (CU i lives on XCD i // 32). ``offset``
-    rotates the starting CU within every XCD so that two co-scheduled pods
-    with disjoint offsets get disjoint masks (the agent allocates offsets).
+    (in CUs, rounded to pairs) rotates the starting pair within every XCD so
+    that two co-scheduled pods with disjoint offsets get disjoint masks.
     """
     per_xcd = total_cus // xcd_count
-    base, extra = divmod(n_cus, xcd_count)
+    pairs_per_xcd = per_xcd // 2
+    n_pairs = (n_cus + 1) // 2
+    pair_offset = offset // 2
+    base, extra = divmod(n_pairs, xcd_count)
     cus: List[int] = []
     for xcd in range(xcd_count):
         take = base + (1 if xcd < extra else 0)
         for k in range(take):
-            cus.append(xcd * per_xcd + (offset + k) % per_xcd)
+            pair = (pair_offset + k) % pairs_per_xcd
+            cu0 = xcd * per_xcd + pair * 2
+            cus.extend((cu0, cu0 + 1))
     return cus
 
 

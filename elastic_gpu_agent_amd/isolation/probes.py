@@ -1,0 +1,94 @@
+"""ctypes wrapper over libegpu_kernels.so (gfx950 verification kernels).
+
+On a GPU box these are the empirical checks behind the isolation layer:
+``census()`` proves a CU mask stuck (distinct CUs observed ≤ mask popcount),
+``throughput_ms()`` proves compute share scales with the mask, and
+``bandwidth_gbps()`` feeds occupancy reporting. Loading fails loudly if the
+extension was not built — there is no eager/Python fallback for these.
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import List, Tuple
+
+_LIB = None
+
+
+def _lib():
+    global _LIB
+    if _LIB is None:
+        path = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                            "libegpu_kernels.so")
+        if not os.path.exists(path):
+            raise RuntimeError(
+                f"{path} not built — run `python -m elastic_gpu_agent_amd.native.build`"
+            )
+        lib = ctypes.CDLL(path)
+        lib.egpu_last_error.restype = ctypes.c_char_p
+        lib.egpu_census.argtypes = [
+            ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.POINTER(ctypes.c_uint32), ctypes.c_int,
+            ctypes.POINTER(ctypes.c_int),
+        ]
+        lib.egpu_throughput.argtypes = [
+            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.POINTER(ctypes.c_float)
+        ]
+        lib.egpu_bandwidth.argtypes = [
+            ctypes.c_int, ctypes.c_int, ctypes.POINTER(ctypes.c_double)
+        ]
+        lib.egpu_device_count.argtypes = [ctypes.POINTER(ctypes.c_int)]
+        lib.egpu_malloc_bytes.argtypes = [ctypes.c_int, ctypes.c_uint64]
+        lib.egpu_free_vram.argtypes = [ctypes.c_int]
+        lib.egpu_free_vram.restype = ctypes.c_uint64
+        _LIB = lib
+    return _LIB
+
+
+class ProbeError(RuntimeError):
+    pass
+
+
+def _check(rc: int, what: str):
+    if rc != 0:
+        raise ProbeError(f"{what} failed (rc={rc}): {_lib().egpu_last_error().decode()}")
+
+
+def device_count() -> int:
+    n = ctypes.c_int(0)
+    _check(_lib().egpu_device_count(ctypes.byref(n)), "egpu_device_count")
+    return n.value
+
+
+def census(device: int = 0, blocks: int = 4096, spin: int = 200000) -> List[int]:
+    """Returns the distinct physical-CU identities ((xcc<<16)|cu bits) that
+    executed at least one workgroup."""
+    max_ids = 1024
+    ids = (ctypes.c_uint32 * max_ids)()
+    n = ctypes.c_int(0)
+    _check(
+        _lib().egpu_census(device, blocks, spin, ids, max_ids, ctypes.byref(n)),
+        "egpu_census",
+    )
+    return list(ids[: n.value])
+
+
+def throughput_ms(device: int = 0, blocks: int = 2048, iters: int = 2_000_000) -> float:
+    ms = ctypes.c_float(0)
+    _check(_lib().egpu_throughput(device, blocks, iters, ctypes.byref(ms)), "egpu_throughput")
+    return ms.value
+
+
+def bandwidth_gbps(device: int = 0, mib: int = 1024) -> float:
+    g = ctypes.c_double(0)
+    _check(_lib().egpu_bandwidth(device, mib, ctypes.byref(g)), "egpu_bandwidth")
+    return g.value
+
+
+def malloc_bytes(device: int, n: int) -> int:
+    """hipMalloc probe; returns hipError_t (0 = success)."""
+    return _lib().egpu_malloc_bytes(device, n)
+
+
+def free_vram(device: int = 0) -> int:
+    return _lib().egpu_free_vram(device)

@@ -13,9 +13,7 @@ import logging
 import threading
 from typing import List, Optional
 
-import grpc
-
-from .. import consts
+from .. import consts, egrpc
 from ..protos import podresources as pr
 from ..types import Device, PodContainer
 
@@ -35,18 +33,12 @@ class KubeletDeviceLocator(DeviceLocator):
         self._resource = resource_name
         self._socket = socket_path
         self._lock = threading.Lock()
-        self._channel: Optional[grpc.Channel] = None
+        self._channel: Optional[egrpc.Channel] = None
         self._list = None
 
     def _ensure(self):
         if self._channel is None:
-            self._channel = grpc.insecure_channel(
-                f"unix://{self._socket}",
-                options=[
-                    ("grpc.max_receive_message_length", consts.POD_RESOURCES_MAX_SIZE),
-                    ("grpc.max_send_message_length", consts.POD_RESOURCES_MAX_SIZE),
-                ],
-            )
+            self._channel = egrpc.Channel(self._socket)
             self._list = self._channel.unary_unary(
                 pr.METHOD_LIST,
                 request_serializer=pr.ListPodResourcesRequest.encode,
@@ -64,7 +56,7 @@ class KubeletDeviceLocator(DeviceLocator):
             self._ensure()
             try:
                 return self._list({}, timeout=10.0)
-            except grpc.RpcError:
+            except egrpc.EgrpcError:
                 # lazy repair: re-dial once (kubelet may have restarted,
                 # ref behavior: pkg/kube/locator.go:47-53)
                 self._reset()

@@ -7,11 +7,9 @@ carries an equivalent server shape in-tree: pkg/podresources/server.go:44-75).
 from __future__ import annotations
 
 import threading
-from concurrent import futures
 from typing import Dict, List
 
-import grpc
-
+from .. import egrpc
 from ..protos import podresources as pr
 
 
@@ -21,19 +19,18 @@ class PodResourcesServer:
         self._lock = threading.Lock()
         # table: (ns, pod) -> {container: [(resource_name, [ids...]), ...]}
         self._table: Dict[tuple, Dict[str, List[tuple]]] = {}
-        handler = grpc.method_handlers_generic_handler(
+        self._server = egrpc.Server()
+        self._server.add_service(
             pr.POD_RESOURCES_SERVICE,
             {
-                "List": grpc.unary_unary_rpc_method_handler(
+                "List": egrpc.unary_unary(
                     self._list,
                     request_deserializer=pr.ListPodResourcesRequest.decode,
                     response_serializer=pr.ListPodResourcesResponse.encode,
                 )
             },
         )
-        self._server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers))
-        self._server.add_generic_rpc_handlers((handler,))
-        self._server.add_insecure_port(f"unix://{socket_path}")
+        self._server.bind_unix(socket_path)
 
     def set_assignment(
         self, namespace: str, pod: str, container: str, resource_name: str, device_ids: List[str]

@@ -1,0 +1,105 @@
+"""In-tree build of every native artifact (gfx950 only, no JIT cache):
+
+  elastic_gpu_agent_amd/_amdsmi<ext>.so   pybind11 → libamd_smi enumeration
+  elastic_gpu_agent_amd/libegpu_shim.so   HSA interposer (CU mask + HBM quota)
+  elastic_gpu_agent_amd/libegpu_kernels.so  gfx950 HIP verification kernels
+  bin/egpu-hook                           OCI prestart hook binary
+
+Run: ``python -m elastic_gpu_agent_amd.native.build`` (hipcc cross-compiles
+without a GPU; artifacts land in-tree so they travel with the repo snapshot).
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import sysconfig
+
+ROCM = os.environ.get("ROCM_PATH", "/opt/rocm")
+HERE = os.path.dirname(os.path.abspath(__file__))
+PKG = os.path.dirname(HERE)
+REPO = os.path.dirname(PKG)
+
+
+def _run(cmd):
+    print("+", " ".join(cmd), flush=True)
+    subprocess.check_call(cmd)
+
+
+def _newer(target: str, *sources: str) -> bool:
+    if not os.path.exists(target):
+        return False
+    t = os.path.getmtime(target)
+    return all(os.path.getmtime(s) <= t for s in sources)
+
+
+def pybind_includes():
+    import pybind11
+
+    return [f"-I{pybind11.get_include()}", f"-I{sysconfig.get_paths()['include']}"]
+
+
+def build_amdsmi(force=False):
+    src = os.path.join(HERE, "amdsmi_binding.cpp")
+    ext = sysconfig.get_config_var("EXT_SUFFIX")
+    out = os.path.join(PKG, f"_amdsmi{ext}")
+    if not force and _newer(out, src):
+        return out
+    _run(
+        ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src]
+        + pybind_includes()
+        + [f"-I{ROCM}/include", f"-L{ROCM}/lib", "-lamd_smi",
+           f"-Wl,-rpath,{ROCM}/lib", "-o", out]
+    )
+    return out
+
+
+def build_shim(force=False):
+    src = os.path.join(HERE, "egpu_shim.cpp")
+    out = os.path.join(PKG, "libegpu_shim.so")
+    if not force and _newer(out, src):
+        return out
+    _run(
+        ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", "-DAMD_INTERNAL_BUILD", src,
+         f"-I{ROCM}/include", f"-I{ROCM}/include/hsa", "-o", out]
+    )
+    return out
+
+
+def build_kernels(force=False):
+    src = os.path.join(HERE, "egpu_kernels.hip")
+    out = os.path.join(PKG, "libegpu_kernels.so")
+    if not force and _newer(out, src):
+        return out
+    hipcc = os.path.join(ROCM, "bin", "hipcc")
+    _run(
+        [hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-shared", "-fPIC", src,
+         "-o", out]
+    )
+    return out
+
+
+def build_hook(force=False):
+    src = os.path.join(HERE, "egpu_hook.cpp")
+    os.makedirs(os.path.join(REPO, "bin"), exist_ok=True)
+    out = os.path.join(REPO, "bin", "egpu-hook")
+    if not force and _newer(out, src):
+        return out
+    _run(["g++", "-O2", "-std=c++17", src, "-o", out])
+    return out
+
+
+def build_all(force=False):
+    return {
+        "amdsmi": build_amdsmi(force),
+        "shim": build_shim(force),
+        "kernels": build_kernels(force),
+        "hook": build_hook(force),
+    }
+
+
+if __name__ == "__main__":
+    force = "--force" in sys.argv
+    arts = build_all(force)
+    for k, v in arts.items():
+        print(f"{k}: {v}")

@@ -1,0 +1,339 @@
+// egpu-hook — OCI prestart hook injecting MI355X device nodes into containers.
+//
+// MI355X-native replacement for the whole host-runtime layer of the
+// reference: its Go hook (cmd/elastic-gpu-hook/main.go) only
+// resolved GPU indexes and then delegated injection to a prebuilt forked
+// nvidia-container-toolkit + nvidia-container-cli. ROCm needs no driver-file
+// injection (userspace lives in the image), so this hook does the whole job
+// itself, first-class:
+//
+//   1. read the OCI hook state JSON from stdin ({pid, bundle});
+//   2. read <bundle>/config.json, find the GPU=<hash> env the agent's
+//      Allocate response set;
+//   3. resolve <dev_root>/elastic-gpu-<hash>-* symlinks (created by the
+//      agent's PreStartContainer) → /dev/dri/renderD<minor> targets;
+//   4. enter the container's mount namespace (setns) and mknod
+//      /dev/kfd + /dev/dri/renderD<minor> with the host major:minor;
+//   5. for cgroup-v1 hosts, whitelist the nodes in the devices cgroup
+//      (v2 grants flow through the kubelet DeviceSpec path instead).
+//
+// No GPU env → passthrough (exit 0), like the reference hook.
+//
+// Env knobs: EGPU_DEV_ROOT (default /dev), EGPU_HOOK_LOG (default
+// /var/log/egpu-hook.log), EGPU_HOOK_DRYRUN=1 (print planned actions as JSON
+// to stdout instead of acting — used by the CPU test-suite).
+
+#include <dirent.h>
+#include <fcntl.h>
+#include <sched.h>
+#include <sys/stat.h>
+#include <sys/sysmacros.h>
+#include <sys/types.h>
+#include <unistd.h>
+
+#include <cstdarg>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+namespace {
+
+FILE* g_log = nullptr;
+
+void logf(const char* fmt, ...) {
+  if (!g_log) return;
+  va_list ap;
+  va_start(ap, fmt);
+  vfprintf(g_log, fmt, ap);
+  fprintf(g_log, "\n");
+  fflush(g_log);
+  va_end(ap);
+}
+
+std::string slurp_stream(FILE* f) {
+  std::string out;
+  char buf[4096];
+  size_t n;
+  while ((n = fread(buf, 1, sizeof(buf), f)) > 0) out.append(buf, n);
+  return out;
+}
+
+std::string slurp_file(const std::string& path) {
+  FILE* f = fopen(path.c_str(), "rb");
+  if (!f) return {};
+  std::string s = slurp_stream(f);
+  fclose(f);
+  return s;
+}
+
+// --- minimal JSON field extraction (OCI state/config are machine-written) ---
+bool find_string(const std::string& body, const char* key, std::string* out) {
+  std::string pat = std::string("\"") + key + "\"";
+  size_t p = body.find(pat);
+  if (p == std::string::npos) return false;
+  p = body.find(':', p + pat.size());
+  if (p == std::string::npos) return false;
+  p = body.find('"', p);
+  if (p == std::string::npos) return false;
+  size_t e = p + 1;
+  std::string s;
+  while (e < body.size() && body[e] != '"') {
+    if (body[e] == '\\' && e + 1 < body.size()) ++e;
+    s += body[e++];
+  }
+  *out = s;
+  return true;
+}
+
+bool find_int(const std::string& body, const char* key, long* out) {
+  std::string pat = std::string("\"") + key + "\"";
+  size_t p = body.find(pat);
+  if (p == std::string::npos) return false;
+  p = body.find(':', p + pat.size());
+  if (p == std::string::npos) return false;
+  ++p;
+  while (p < body.size() && (body[p] == ' ' || body[p] == '\t')) ++p;
+  char* end = nullptr;
+  long v = strtol(body.c_str() + p, &end, 10);
+  if (end == body.c_str() + p) return false;
+  *out = v;
+  return true;
+}
+
+// extract the "env" string array from an OCI config.json
+std::vector<std::string> find_env(const std::string& body) {
+  std::vector<std::string> envs;
+  size_t p = body.find("\"env\"");
+  if (p == std::string::npos) return envs;
+  p = body.find('[', p);
+  if (p == std::string::npos) return envs;
+  size_t end = body.find(']', p);
+  if (end == std::string::npos) return envs;
+  size_t q = p;
+  while (true) {
+    q = body.find('"', q + 1);
+    if (q == std::string::npos || q > end) break;
+    size_t e = q + 1;
+    std::string s;
+    while (e < body.size() && body[e] != '"') {
+      if (body[e] == '\\' && e + 1 < body.size()) ++e;
+      s += body[e++];
+    }
+    envs.push_back(s);
+    q = e;
+  }
+  return envs;
+}
+
+std::string env_value(const std::vector<std::string>& envs, const std::string& key) {
+  for (const auto& e : envs) {
+    if (e.size() > key.size() + 1 && e.compare(0, key.size(), key) == 0 &&
+        e[key.size()] == '=') {
+      return e.substr(key.size() + 1);
+    }
+  }
+  return {};
+}
+
+struct DeviceNode {
+  std::string path;  // canonical in-container path
+  unsigned maj = 0, min = 0;
+};
+
+// resolve elastic-gpu-<hash>-* links under dev_root → render-node targets
+std::vector<std::string> resolve_gpu_links(const std::string& dev_root,
+                                           const std::string& hash) {
+  std::vector<std::string> targets;
+  std::string prefix = "elastic-gpu-" + hash + "-";
+  DIR* d = opendir(dev_root.c_str());
+  if (!d) return targets;
+  struct dirent* ent;
+  while ((ent = readdir(d)) != nullptr) {
+    std::string name = ent->d_name;
+    if (name.rfind(prefix, 0) != 0) continue;
+    std::string link = dev_root + "/" + name;
+    char buf[512];
+    ssize_t n = readlink(link.c_str(), buf, sizeof(buf) - 1);
+    if (n <= 0) continue;
+    buf[n] = '\0';
+    targets.push_back(buf);
+  }
+  closedir(d);
+  return targets;
+}
+
+bool stat_node(const std::string& path, unsigned* maj, unsigned* min) {
+  struct stat st {};
+  if (stat(path.c_str(), &st) != 0 || !S_ISCHR(st.st_mode)) return false;
+  *maj = major(st.st_rdev);
+  *min = minor(st.st_rdev);
+  return true;
+}
+
+// cgroup-v1 devices controller path of a pid ("" when v2/absent)
+std::string devices_cgroup_path(long pid) {
+  std::string body = slurp_file("/proc/" + std::to_string(pid) + "/cgroup");
+  size_t pos = 0;
+  while (pos < body.size()) {
+    size_t eol = body.find('\n', pos);
+    std::string line = body.substr(pos, eol == std::string::npos ? body.size() - pos : eol - pos);
+    pos = (eol == std::string::npos) ? body.size() : eol + 1;
+    // format: N:controllers:path
+    size_t c1 = line.find(':');
+    size_t c2 = line.find(':', c1 + 1);
+    if (c1 == std::string::npos || c2 == std::string::npos) continue;
+    std::string controllers = line.substr(c1 + 1, c2 - c1 - 1);
+    if (controllers.find("devices") != std::string::npos)
+      return "/sys/fs/cgroup/devices" + line.substr(c2 + 1);
+  }
+  return {};
+}
+
+int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun) {
+  if (dryrun) {
+    printf("{\"pid\": %ld, \"nodes\": [", pid);
+    for (size_t i = 0; i < nodes.size(); ++i) {
+      printf("%s{\"path\": \"%s\", \"major\": %u, \"minor\": %u}", i ? ", " : "",
+             nodes[i].path.c_str(), nodes[i].maj, nodes[i].min);
+    }
+    printf("]}\n");
+    return 0;
+  }
+
+  // cgroup v1 allow-list (v2 device access was granted via kubelet DeviceSpec)
+  std::string cg = devices_cgroup_path(pid);
+  if (!cg.empty()) {
+    std::string allow = cg + "/devices.allow";
+    FILE* f = fopen(allow.c_str(), "w");
+    if (f) {
+      for (const auto& n : nodes) {
+        fprintf(f, "c %u:%u rwm", n.maj, n.min);
+        fflush(f);
+      }
+      fclose(f);
+      logf("cgroup v1 allow-listed %zu nodes in %s", nodes.size(), cg.c_str());
+    } else {
+      logf("WARN: cannot open %s: %s", allow.c_str(), strerror(errno));
+    }
+  }
+
+  // enter the container's mount namespace and create the nodes
+  std::string nspath = "/proc/" + std::to_string(pid) + "/ns/mnt";
+  int fd = open(nspath.c_str(), O_RDONLY);
+  if (fd < 0) {
+    logf("ERROR: open %s: %s", nspath.c_str(), strerror(errno));
+    return 1;
+  }
+  if (setns(fd, CLONE_NEWNS) != 0) {
+    logf("ERROR: setns: %s", strerror(errno));
+    close(fd);
+    return 1;
+  }
+  close(fd);
+
+  for (const auto& n : nodes) {
+    // ensure parent dir
+    size_t slash = n.path.rfind('/');
+    if (slash != std::string::npos) {
+      std::string dir = n.path.substr(0, slash);
+      for (size_t i = 1; i < dir.size(); ++i) {
+        if (dir[i] == '/') mkdir(dir.substr(0, i).c_str(), 0755);
+      }
+      mkdir(dir.c_str(), 0755);
+    }
+    struct stat st {};
+    if (stat(n.path.c_str(), &st) == 0) continue;  // already present
+    if (mknod(n.path.c_str(), S_IFCHR | 0666, makedev(n.maj, n.min)) != 0) {
+      logf("ERROR: mknod %s c %u:%u: %s", n.path.c_str(), n.maj, n.min, strerror(errno));
+      return 1;
+    }
+    chmod(n.path.c_str(), 0666);
+    logf("mknod %s c %u:%u", n.path.c_str(), n.maj, n.min);
+  }
+  return 0;
+}
+
+}  // namespace
+
+int main(int argc, char** argv) {
+  // accept NVIDIA-hook-style lifecycle argument; only prestart acts
+  if (argc > 1 && strcmp(argv[1], "prestart") != 0 && strcmp(argv[1], "createRuntime") != 0) {
+    return 0;
+  }
+  const char* logpath = getenv("EGPU_HOOK_LOG");
+  g_log = fopen(logpath ? logpath : "/var/log/egpu-hook.log", "a");
+  bool dryrun = getenv("EGPU_HOOK_DRYRUN") && getenv("EGPU_HOOK_DRYRUN")[0] == '1';
+  const char* dev_root_env = getenv("EGPU_DEV_ROOT");
+  std::string dev_root = dev_root_env ? dev_root_env : "/dev";
+
+  std::string state = slurp_stream(stdin);
+  long pid = 0;
+  std::string bundle;
+  if (!find_int(state, "pid", &pid) || !find_string(state, "bundle", &bundle)) {
+    // tolerate annotations-style "bundlePath"
+    if (!find_string(state, "bundlePath", &bundle) || pid == 0) {
+      logf("ERROR: bad hook state: %s", state.c_str());
+      return 1;
+    }
+  }
+
+  std::string config = slurp_file(bundle + "/config.json");
+  if (config.empty()) {
+    logf("ERROR: cannot read %s/config.json", bundle.c_str());
+    return 1;
+  }
+  auto envs = find_env(config);
+  std::string hash = env_value(envs, "GPU");
+  if (hash.empty()) {
+    logf("no GPU env; passthrough (pid %ld)", pid);
+    return 0;  // not an elastic-gpu container
+  }
+
+  auto targets = resolve_gpu_links(dev_root, hash);
+  if (targets.empty()) {
+    logf("ERROR: no elastic-gpu-%s-* links under %s", hash.c_str(), dev_root.c_str());
+    return 1;
+  }
+
+  std::vector<DeviceNode> nodes;
+  DeviceNode kfd;
+  kfd.path = "/dev/kfd";
+  if (!stat_node(dev_root + "/kfd", &kfd.maj, &kfd.min)) {
+    if (dryrun) {  // CPU test boxes have no /dev/kfd: use a placeholder
+      kfd.maj = 234;
+      kfd.min = 0;
+    } else {
+      logf("ERROR: no /dev/kfd on host");
+      return 1;
+    }
+  }
+  nodes.push_back(kfd);
+  for (const auto& t : targets) {
+    DeviceNode n;
+    // target is /dev/dri/renderD<minor>; canonical path inside the container
+    n.path = t;
+    std::string host_path = t;
+    if (t.rfind("/dev/", 0) == 0 && dev_root != "/dev")
+      host_path = dev_root + t.substr(4);  // test roots: /dev/x → <root>/x
+    unsigned maj = 226, min = 0;
+    if (!stat_node(host_path, &maj, &min)) {
+      // derive the minor from the renderD<N> name (DRM render major = 226)
+      size_t rp = t.rfind("renderD");
+      if (rp == std::string::npos) {
+        logf("ERROR: unexpected link target %s", t.c_str());
+        return 1;
+      }
+      maj = 226;
+      min = (unsigned)atoi(t.c_str() + rp + 7);
+    }
+    n.maj = maj;
+    n.min = min;
+    nodes.push_back(n);
+  }
+
+  logf("injecting %zu nodes for GPU=%s into pid %ld%s", nodes.size(), hash.c_str(), pid,
+       dryrun ? " (dry-run)" : "");
+  return inject(pid, nodes, dryrun);
+}

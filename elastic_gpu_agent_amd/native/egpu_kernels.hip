@@ -1,0 +1,192 @@
+// libegpu_kernels.so — gfx950 verification kernels for the isolation layer.
+//
+// Three probes, all with a C ABI (loaded via ctypes by the agent's tests and
+// the occupancy reporter):
+//
+//   egpu_census      — every workgroup records its physical CU identity
+//                      (HW_REG_XCC_ID + HW_REG_HW_ID) so the host can count
+//                      the DISTINCT CUs a masked queue actually dispatched
+//                      to: the empirical check that a CU mask stuck.
+//   egpu_throughput  — fixed FP32 FMA workload; a pod confined to X% of the
+//                      CUs should land at ≈X% of the full-card rate.
+//   egpu_bandwidth   — float4 streaming read (the HBM3E-bound shape from
+//                      the CDNA4 guide) for occupancy/QoS reporting.
+//   egpu_malloc_bytes— hipMalloc probe used by the HBM-quota tests.
+//
+// CDNA4 notes (see /opt/skills/guides/cdna_hip_programming.md): wave=64,
+// blocks are multiples of 64 threads; the census spin keeps blocks resident
+// long enough that the dispatcher touches every enabled CU; grids are ≫256
+// workgroups to fill all 8 XCDs.
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+
+#define EGPU_CHECK(expr)                                   \
+  do {                                                     \
+    hipError_t _e = (expr);                                \
+    if (_e != hipSuccess) {                                \
+      snprintf(g_err, sizeof(g_err), "%s: %s", #expr,      \
+               hipGetErrorString(_e));                     \
+      return (int)_e;                                      \
+    }                                                      \
+  } while (0)
+
+static char g_err[512] = {0};
+
+extern "C" const char* egpu_last_error() { return g_err; }
+
+// ---------------------------------------------------------------- census
+
+__global__ void __launch_bounds__(256) census_kernel(uint32_t* out, int spin) {
+  // identity of the CU this workgroup landed on
+  uint32_t hwid, xcc;
+  asm volatile("s_getreg_b32 %0, hwreg(HW_REG_HW_ID)" : "=s"(hwid));
+  asm volatile("s_getreg_b32 %0, hwreg(HW_REG_XCC_ID)" : "=s"(xcc));
+  // keep the block resident for a while so co-resident blocks spread over
+  // every enabled CU instead of a few fast-retiring ones
+  float acc = threadIdx.x * 1.0f;
+  for (int i = 0; i < spin; ++i) acc = fmaf(acc, 1.0000001f, 0.5f);
+  if (threadIdx.x == 0) {
+    // bits [14:8] of HW_ID = {SE_ID[14:13], SH_ID[12], CU_ID[11:8]} on gfx9
+    // lineage; XCC_ID identifies the XCD. Together: one physical CU.
+    out[blockIdx.x] = (xcc << 16) | ((hwid >> 8) & 0x7F) | ((acc > 1e30f) ? 0 : 0);
+  }
+}
+
+// Returns 0 on success; *n_distinct = number of distinct CUs observed;
+// cu_ids[0..*n_distinct) = their (xcc<<16 | cu) identities.
+extern "C" int egpu_census(int device, int blocks, int spin, uint32_t* cu_ids, int max_ids,
+                           int* n_distinct) {
+  EGPU_CHECK(hipSetDevice(device));
+  uint32_t* d_out = nullptr;
+  EGPU_CHECK(hipMalloc(&d_out, blocks * sizeof(uint32_t)));
+  EGPU_CHECK(hipMemset(d_out, 0xFF, blocks * sizeof(uint32_t)));
+  hipLaunchKernelGGL(census_kernel, dim3(blocks), dim3(256), 0, 0, d_out, spin);
+  EGPU_CHECK(hipGetLastError());
+  EGPU_CHECK(hipDeviceSynchronize());
+  uint32_t* h_out = new uint32_t[blocks];
+  EGPU_CHECK(hipMemcpy(h_out, d_out, blocks * sizeof(uint32_t), hipMemcpyDeviceToHost));
+  EGPU_CHECK(hipFree(d_out));
+  int n = 0;
+  for (int i = 0; i < blocks; ++i) {
+    uint32_t id = h_out[i];
+    if (id == 0xFFFFFFFFu) continue;
+    bool seen = false;
+    for (int j = 0; j < n; ++j)
+      if (cu_ids[j] == id) {
+        seen = true;
+        break;
+      }
+    if (!seen && n < max_ids) cu_ids[n++] = id;
+  }
+  delete[] h_out;
+  *n_distinct = n;
+  return 0;
+}
+
+// ---------------------------------------------------------------- throughput
+
+__global__ void __launch_bounds__(256) fma_kernel(float* out, int iters) {
+  float a = threadIdx.x * 0.001f + 1.0f;
+  float b = blockIdx.x * 0.001f + 1.0f;
+  float c = 0.0f, d = 1.0f;
+  for (int i = 0; i < iters; ++i) {
+    // two independent chains for ILP
+    c = fmaf(a, b, c);
+    d = fmaf(a, 1.0000001f, d);
+  }
+  if (c + d > 1e30f) out[blockIdx.x * blockDim.x + threadIdx.x] = c + d;
+}
+
+// Fixed FMA workload; *ms = wall time. Throughput should scale ≈ linearly
+// with the number of enabled CUs.
+extern "C" int egpu_throughput(int device, int blocks, int iters, float* ms) {
+  EGPU_CHECK(hipSetDevice(device));
+  float* d_out = nullptr;
+  EGPU_CHECK(hipMalloc(&d_out, blocks * 256 * sizeof(float)));
+  hipEvent_t t0, t1;
+  EGPU_CHECK(hipEventCreate(&t0));
+  EGPU_CHECK(hipEventCreate(&t1));
+  // warmup
+  hipLaunchKernelGGL(fma_kernel, dim3(blocks), dim3(256), 0, 0, d_out, iters / 10);
+  EGPU_CHECK(hipDeviceSynchronize());
+  EGPU_CHECK(hipEventRecord(t0, 0));
+  hipLaunchKernelGGL(fma_kernel, dim3(blocks), dim3(256), 0, 0, d_out, iters);
+  EGPU_CHECK(hipEventRecord(t1, 0));
+  EGPU_CHECK(hipEventSynchronize(t1));
+  EGPU_CHECK(hipEventElapsedTime(ms, t0, t1));
+  EGPU_CHECK(hipEventDestroy(t0));
+  EGPU_CHECK(hipEventDestroy(t1));
+  EGPU_CHECK(hipFree(d_out));
+  return 0;
+}
+
+// ---------------------------------------------------------------- bandwidth
+
+__global__ void __launch_bounds__(256) copy_kernel(const float4* __restrict__ src,
+                                                   float4* __restrict__ dst, size_t n4) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < n4; i += stride) dst[i] = src[i];
+}
+
+// Streaming float4 copy of `mib` MiB; *gbps = achieved GB/s (read+write).
+extern "C" int egpu_bandwidth(int device, int mib, double* gbps) {
+  EGPU_CHECK(hipSetDevice(device));
+  size_t bytes = (size_t)mib * 1024 * 1024;
+  size_t n4 = bytes / sizeof(float4);
+  float4 *d_src = nullptr, *d_dst = nullptr;
+  EGPU_CHECK(hipMalloc(&d_src, bytes));
+  EGPU_CHECK(hipMalloc(&d_dst, bytes));
+  EGPU_CHECK(hipMemset(d_src, 1, bytes));
+  int blocks = 4096;  // ≫256 WGs: fill all 8 XCDs
+  hipEvent_t t0, t1;
+  EGPU_CHECK(hipEventCreate(&t0));
+  EGPU_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(copy_kernel, dim3(blocks), dim3(256), 0, 0, d_src, d_dst, n4);  // warm
+  EGPU_CHECK(hipDeviceSynchronize());
+  EGPU_CHECK(hipEventRecord(t0, 0));
+  hipLaunchKernelGGL(copy_kernel, dim3(blocks), dim3(256), 0, 0, d_src, d_dst, n4);
+  EGPU_CHECK(hipEventRecord(t1, 0));
+  EGPU_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  EGPU_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  *gbps = (2.0 * bytes / 1e9) / (ms / 1e3);
+  EGPU_CHECK(hipEventDestroy(t0));
+  EGPU_CHECK(hipEventDestroy(t1));
+  EGPU_CHECK(hipFree(d_src));
+  EGPU_CHECK(hipFree(d_dst));
+  return 0;
+}
+
+// ---------------------------------------------------------------- misc
+
+extern "C" int egpu_device_count(int* count) {
+  EGPU_CHECK(hipGetDeviceCount(count));
+  return 0;
+}
+
+// hipMalloc probe for the HBM-quota tests: returns the hipError_t (0 = the
+// allocation succeeded and was freed).
+extern "C" int egpu_malloc_bytes(int device, uint64_t bytes) {
+  EGPU_CHECK(hipSetDevice(device));
+  void* p = nullptr;
+  hipError_t e = hipMalloc(&p, (size_t)bytes);
+  if (e == hipSuccess) {
+    (void)hipFree(p);
+    return 0;
+  }
+  snprintf(g_err, sizeof(g_err), "hipMalloc(%llu): %s", (unsigned long long)bytes,
+           hipGetErrorString(e));
+  return (int)e;
+}
+
+extern "C" uint64_t egpu_free_vram(int device) {
+  if (hipSetDevice(device) != hipSuccess) return 0;
+  size_t free_b = 0, total_b = 0;
+  if (hipMemGetInfo(&free_b, &total_b) != hipSuccess) return 0;
+  return free_b;
+}

@@ -1,0 +1,381 @@
+// libegpu_shim.so — hand-written HSA interposer enforcing per-container
+// compute (CU mask) and memory (HBM quota) limits on MI355X (gfx950).
+//
+// This is the MI355X re-implementation of the QoS layer the reference
+// delegated to a closed-source driver (SURVEY §7 step 7 — the OSS reference
+// only scopes device nodes). It is loaded into *workload* containers via
+// HSA_TOOLS_LIB (set by the agent's Allocate response): ROCr dlopens this
+// library during hsa_init and calls OnLoad() with its API table; we wrap:
+//
+//   hsa_queue_create            → apply the allocation's CU mask to every new
+//                                 queue (hsa_amd_queue_cu_set_mask);
+//   hsa_amd_queue_cu_set_mask   → intersect caller masks with the allocation
+//                                 mask (a container cannot widen itself);
+//   hsa_amd_memory_pool_allocate/ free,
+//   hsa_memory_allocate / free  → account device-local (VRAM) bytes against
+//                                 the HBM quota; exceeding it fails with
+//                                 HSA_STATUS_ERROR_OUT_OF_RESOURCES, which
+//                                 HIP surfaces as hipErrorOutOfMemory.
+//
+// Limits come from the files the agent mounts at /etc/egpu/limits-*.json
+// (EGPU_LIMITS_DIR overrides the directory) or from env overrides
+// EGPU_CU_MASK (comma-separated LE hex words) / EGPU_MEM_LIMIT_BYTES.
+// The mask uses CU-pair granularity as ROCr requires (hsa_ext_amd.h).
+
+#include <hsa/hsa.h>
+#include <hsa/hsa_api_trace.h>
+#include <hsa/hsa_ext_amd.h>
+
+#include <atomic>
+#include <cstdarg>
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <dirent.h>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+#include <unordered_set>
+#include <vector>
+
+namespace {
+
+// ---------------------------------------------------------------- logging
+bool g_verbose = false;
+
+void logf(const char* fmt, ...) {
+  if (!g_verbose) return;
+  va_list ap;
+  va_start(ap, fmt);
+  fprintf(stderr, "[egpu-shim] ");
+  vfprintf(stderr, fmt, ap);
+  fprintf(stderr, "\n");
+  va_end(ap);
+}
+
+// ---------------------------------------------------------------- config
+constexpr int kMaskWordsMax = 16;  // up to 512 CUs
+
+struct Config {
+  bool have_mask = false;
+  uint32_t mask[kMaskWordsMax] = {0};
+  int mask_words = 0;
+  uint64_t mem_limit = 0;  // 0 = unlimited
+};
+
+Config g_cfg;
+
+// minimal extraction from the agent-generated limits JSON (flat, trusted
+// producer): finds "key": <string|number> at top level.
+bool json_find_string(const std::string& body, const char* key, std::string* out) {
+  std::string pat = std::string("\"") + key + "\"";
+  size_t p = body.find(pat);
+  if (p == std::string::npos) return false;
+  p = body.find(':', p + pat.size());
+  if (p == std::string::npos) return false;
+  p = body.find('"', p);
+  if (p == std::string::npos) return false;
+  size_t e = body.find('"', p + 1);
+  if (e == std::string::npos) return false;
+  *out = body.substr(p + 1, e - p - 1);
+  return true;
+}
+
+bool json_find_u64(const std::string& body, const char* key, uint64_t* out) {
+  std::string pat = std::string("\"") + key + "\"";
+  size_t p = body.find(pat);
+  if (p == std::string::npos) return false;
+  p = body.find(':', p + pat.size());
+  if (p == std::string::npos) return false;
+  ++p;
+  while (p < body.size() && (body[p] == ' ' || body[p] == '\t')) ++p;
+  char* end = nullptr;
+  unsigned long long v = strtoull(body.c_str() + p, &end, 10);
+  if (end == body.c_str() + p) return false;
+  *out = v;
+  return true;
+}
+
+bool parse_mask_hex(const char* s, Config* cfg) {
+  int w = 0;
+  const char* p = s;
+  while (*p && w < kMaskWordsMax) {
+    char* end = nullptr;
+    unsigned long v = strtoul(p, &end, 16);
+    if (end == p) return false;
+    cfg->mask[w++] = static_cast<uint32_t>(v);
+    p = (*end == ',') ? end + 1 : end;
+    if (*end == '\0') break;
+  }
+  if (w == 0) return false;
+  cfg->mask_words = w;
+  // all-ones masks are equivalent to no mask
+  bool all_ones = true;
+  for (int i = 0; i < w; ++i)
+    if (cfg->mask[i] != 0xFFFFFFFFu) all_ones = false;
+  cfg->have_mask = !all_ones;
+  return true;
+}
+
+std::string slurp(const std::string& path) {
+  FILE* f = fopen(path.c_str(), "rb");
+  if (!f) return {};
+  std::string out;
+  char buf[4096];
+  size_t n;
+  while ((n = fread(buf, 1, sizeof(buf), f)) > 0) out.append(buf, n);
+  fclose(f);
+  return out;
+}
+
+void load_config() {
+  const char* verbose = getenv("EGPU_SHIM_VERBOSE");
+  g_verbose = verbose && verbose[0] == '1';
+
+  const char* dir = getenv("EGPU_LIMITS_DIR");
+  std::string limits_dir = dir ? dir : "/etc/egpu";
+  DIR* d = opendir(limits_dir.c_str());
+  if (d) {
+    struct dirent* ent;
+    while ((ent = readdir(d)) != nullptr) {
+      std::string name = ent->d_name;
+      if (name.rfind("limits", 0) != 0 || name.size() < 5 ||
+          name.substr(name.size() - 5) != ".json")
+        continue;
+      std::string body = slurp(limits_dir + "/" + name);
+      if (body.empty()) continue;
+      std::string mask;
+      if (json_find_string(body, "cu_mask", &mask)) parse_mask_hex(mask.c_str(), &g_cfg);
+      uint64_t mem = 0;
+      if (json_find_u64(body, "mem_limit_bytes", &mem) && mem > 0) g_cfg.mem_limit = mem;
+      logf("loaded %s (mask=%d mem_limit=%llu)", name.c_str(), g_cfg.have_mask ? 1 : 0,
+           (unsigned long long)g_cfg.mem_limit);
+    }
+    closedir(d);
+  }
+  // env overrides (tests / manual runs)
+  if (const char* m = getenv("EGPU_CU_MASK")) parse_mask_hex(m, &g_cfg);
+  if (const char* l = getenv("EGPU_MEM_LIMIT_BYTES")) g_cfg.mem_limit = strtoull(l, nullptr, 10);
+  logf("config: have_mask=%d words=%d mem_limit=%llu", g_cfg.have_mask, g_cfg.mask_words,
+       (unsigned long long)g_cfg.mem_limit);
+}
+
+// ---------------------------------------------------------------- state
+CoreApiTable g_core{};     // saved original core entry points
+AmdExtTable g_amdext{};    // saved original amd-ext entry points
+std::atomic<uint64_t> g_vram_used{0};
+std::atomic<uint64_t> g_denied{0};
+std::atomic<uint64_t> g_queues_masked{0};
+std::mutex g_alloc_mu;
+std::unordered_map<void*, uint64_t> g_alloc_sizes;
+
+// VRAM pool/region classification (lazily built: agents exist only after
+// hsa_init completes, long before the first allocation).
+std::once_flag g_pools_once;
+std::unordered_set<uint64_t> g_vram_pools;    // hsa_amd_memory_pool_t.handle
+std::unordered_set<uint64_t> g_vram_regions;  // hsa_region_t.handle
+
+void build_pool_sets() {
+  auto agent_cb = [](hsa_agent_t agent, void*) -> hsa_status_t {
+    hsa_device_type_t dev{};
+    if (g_core.hsa_agent_get_info_fn(agent, HSA_AGENT_INFO_DEVICE, &dev) != HSA_STATUS_SUCCESS ||
+        dev != HSA_DEVICE_TYPE_GPU)
+      return HSA_STATUS_SUCCESS;
+    auto pool_cb = [](hsa_amd_memory_pool_t pool, void*) -> hsa_status_t {
+      hsa_amd_segment_t seg{};
+      if (g_amdext.hsa_amd_memory_pool_get_info_fn(pool, HSA_AMD_MEMORY_POOL_INFO_SEGMENT, &seg) ==
+              HSA_STATUS_SUCCESS &&
+          seg == HSA_AMD_SEGMENT_GLOBAL) {
+        g_vram_pools.insert(pool.handle);
+      }
+      return HSA_STATUS_SUCCESS;
+    };
+    g_amdext.hsa_amd_agent_iterate_memory_pools_fn(agent, pool_cb, nullptr);
+    auto region_cb = [](hsa_region_t region, void*) -> hsa_status_t {
+      hsa_region_segment_t seg{};
+      if (g_core.hsa_region_get_info_fn(region, HSA_REGION_INFO_SEGMENT, &seg) ==
+              HSA_STATUS_SUCCESS &&
+          seg == HSA_REGION_SEGMENT_GLOBAL) {
+        g_vram_regions.insert(region.handle);
+      }
+      return HSA_STATUS_SUCCESS;
+    };
+    g_core.hsa_agent_iterate_regions_fn(agent, region_cb, nullptr);
+    return HSA_STATUS_SUCCESS;
+  };
+  g_core.hsa_iterate_agents_fn(agent_cb, nullptr);
+  logf("classified %zu VRAM pools, %zu VRAM regions", g_vram_pools.size(),
+       g_vram_regions.size());
+}
+
+bool is_vram_pool(hsa_amd_memory_pool_t pool) {
+  std::call_once(g_pools_once, build_pool_sets);
+  return g_vram_pools.count(pool.handle) != 0;
+}
+
+bool is_vram_region(hsa_region_t region) {
+  std::call_once(g_pools_once, build_pool_sets);
+  return g_vram_regions.count(region.handle) != 0;
+}
+
+void record_ptr(void* ptr, uint64_t size) {
+  if (g_cfg.mem_limit == 0) return;
+  std::lock_guard<std::mutex> lk(g_alloc_mu);
+  g_alloc_sizes[ptr] = size;
+}
+
+void refund(void* ptr) {
+  if (g_cfg.mem_limit == 0 || ptr == nullptr) return;
+  std::lock_guard<std::mutex> lk(g_alloc_mu);
+  auto it = g_alloc_sizes.find(ptr);
+  if (it != g_alloc_sizes.end()) {
+    g_vram_used.fetch_sub(it->second);
+    g_alloc_sizes.erase(it);
+  }
+}
+
+// ---------------------------------------------------------------- wrappers
+void apply_mask(const hsa_queue_t* queue, hsa_agent_t agent) {
+  if (!g_cfg.have_mask) return;
+  uint32_t cu_count = 0;
+  if (g_core.hsa_agent_get_info_fn(
+          agent, static_cast<hsa_agent_info_t>(HSA_AMD_AGENT_INFO_COMPUTE_UNIT_COUNT),
+          &cu_count) != HSA_STATUS_SUCCESS)
+    cu_count = 32u * g_cfg.mask_words;
+  uint32_t bits = ((cu_count + 31) / 32) * 32;
+  if (bits > 32u * kMaskWordsMax) bits = 32u * kMaskWordsMax;
+  // pad mask to the agent's width with zeros (extra CUs stay disabled)
+  uint32_t words[kMaskWordsMax] = {0};
+  for (int i = 0; i < g_cfg.mask_words && i < kMaskWordsMax; ++i) words[i] = g_cfg.mask[i];
+  hsa_status_t st = g_amdext.hsa_amd_queue_cu_set_mask_fn(queue, bits, words);
+  if (st == HSA_STATUS_SUCCESS) {
+    g_queues_masked.fetch_add(1);
+    logf("applied CU mask (%u bits) to queue %p", bits, (const void*)queue);
+  } else {
+    fprintf(stderr, "[egpu-shim] ERROR: cu_set_mask failed (%d) — compute limit NOT applied\n",
+            (int)st);
+  }
+}
+
+hsa_status_t queue_create_wrap(hsa_agent_t agent, uint32_t size, hsa_queue_type32_t type,
+                               void (*callback)(hsa_status_t, hsa_queue_t*, void*), void* data,
+                               uint32_t private_segment_size, uint32_t group_segment_size,
+                               hsa_queue_t** queue) {
+  hsa_status_t st = g_core.hsa_queue_create_fn(agent, size, type, callback, data,
+                                               private_segment_size, group_segment_size, queue);
+  if (st == HSA_STATUS_SUCCESS && queue && *queue) apply_mask(*queue, agent);
+  return st;
+}
+
+hsa_status_t cu_set_mask_wrap(const hsa_queue_t* queue, uint32_t num_cu_mask_count,
+                              const uint32_t* cu_mask) {
+  if (!g_cfg.have_mask)
+    return g_amdext.hsa_amd_queue_cu_set_mask_fn(queue, num_cu_mask_count, cu_mask);
+  // intersect the caller's request with the allocation mask: a container may
+  // narrow its own queues but never widen past its quota
+  uint32_t words[kMaskWordsMax] = {0};
+  uint32_t n = num_cu_mask_count / 32;
+  if (n > kMaskWordsMax) n = kMaskWordsMax;
+  for (uint32_t i = 0; i < n; ++i) {
+    uint32_t ours = (i < (uint32_t)g_cfg.mask_words) ? g_cfg.mask[i] : 0;
+    words[i] = (cu_mask ? cu_mask[i] : 0) & ours;
+  }
+  return g_amdext.hsa_amd_queue_cu_set_mask_fn(queue, n * 32, words);
+}
+
+hsa_status_t pool_allocate_wrap(hsa_amd_memory_pool_t pool, size_t size, uint32_t flags,
+                                void** ptr) {
+  if (g_cfg.mem_limit != 0 && is_vram_pool(pool)) {
+    // reserve first (ptr unknown yet): optimistic charge on a sentinel,
+    // re-keyed after the real pointer exists
+    uint64_t prev = g_vram_used.fetch_add(size);
+    if (prev + size > g_cfg.mem_limit) {
+      g_vram_used.fetch_sub(size);
+      g_denied.fetch_add(1);
+      logf("DENY pool alloc %zu (used %llu / limit %llu)", size, (unsigned long long)prev,
+           (unsigned long long)g_cfg.mem_limit);
+      return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
+    }
+    hsa_status_t st = g_amdext.hsa_amd_memory_pool_allocate_fn(pool, size, flags, ptr);
+    if (st != HSA_STATUS_SUCCESS) {
+      g_vram_used.fetch_sub(size);
+      return st;
+    }
+    record_ptr(*ptr, size);
+    return st;
+  }
+  return g_amdext.hsa_amd_memory_pool_allocate_fn(pool, size, flags, ptr);
+}
+
+hsa_status_t pool_free_wrap(void* ptr) {
+  refund(ptr);
+  return g_amdext.hsa_amd_memory_pool_free_fn(ptr);
+}
+
+hsa_status_t memory_allocate_wrap(hsa_region_t region, size_t size, void** ptr) {
+  if (g_cfg.mem_limit != 0 && is_vram_region(region)) {
+    uint64_t prev = g_vram_used.fetch_add(size);
+    if (prev + size > g_cfg.mem_limit) {
+      g_vram_used.fetch_sub(size);
+      g_denied.fetch_add(1);
+      return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
+    }
+    hsa_status_t st = g_core.hsa_memory_allocate_fn(region, size, ptr);
+    if (st != HSA_STATUS_SUCCESS) {
+      g_vram_used.fetch_sub(size);
+      return st;
+    }
+    record_ptr(*ptr, size);
+    return st;
+  }
+  return g_core.hsa_memory_allocate_fn(region, size, ptr);
+}
+
+hsa_status_t memory_free_wrap(void* ptr) {
+  refund(ptr);
+  return g_core.hsa_memory_free_fn(ptr);
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------- tool entry
+extern "C" {
+
+// Introspection API (tests / occupancy reporting dlsym these).
+bool egpu_shim_active() { return g_cfg.have_mask || g_cfg.mem_limit != 0; }
+uint64_t egpu_shim_vram_used() { return g_vram_used.load(); }
+uint64_t egpu_shim_mem_limit() { return g_cfg.mem_limit; }
+uint64_t egpu_shim_denied_allocs() { return g_denied.load(); }
+uint64_t egpu_shim_queues_masked() { return g_queues_masked.load(); }
+int egpu_shim_cu_mask(uint32_t* out, int max_words) {
+  if (!g_cfg.have_mask) return 0;
+  int n = g_cfg.mask_words < max_words ? g_cfg.mask_words : max_words;
+  for (int i = 0; i < n; ++i) out[i] = g_cfg.mask[i];
+  return n;
+}
+
+bool OnLoad(void* table_ptr, uint64_t runtime_version, uint64_t failed_tool_count,
+            const char* const* failed_tool_names) {
+  (void)runtime_version;
+  (void)failed_tool_count;
+  (void)failed_tool_names;
+  load_config();
+  auto* table = reinterpret_cast<HsaApiTable*>(table_ptr);
+  // save originals, then patch the dispatch table
+  memcpy(&g_core, table->core_, sizeof(CoreApiTable));
+  memcpy(&g_amdext, table->amd_ext_, sizeof(AmdExtTable));
+  table->core_->hsa_queue_create_fn = queue_create_wrap;
+  table->amd_ext_->hsa_amd_queue_cu_set_mask_fn = cu_set_mask_wrap;
+  table->amd_ext_->hsa_amd_memory_pool_allocate_fn = pool_allocate_wrap;
+  table->amd_ext_->hsa_amd_memory_pool_free_fn = pool_free_wrap;
+  table->core_->hsa_memory_allocate_fn = memory_allocate_wrap;
+  table->core_->hsa_memory_free_fn = memory_free_wrap;
+  logf("installed (mask=%d mem_limit=%llu)", g_cfg.have_mask,
+       (unsigned long long)g_cfg.mem_limit);
+  return true;
+}
+
+void OnUnload() {}
+
+}  // extern "C"

@@ -8,9 +8,12 @@ re-serve + re-register after a kubelet restart. The fsnotify dependency is
 replaced by inode stat-polling (no extra packages; 1 s period like the
 reference's debounce).
 
-gRPC handlers are wired through generic method handlers with our wire codec
-as (de)serializers — request bytes go straight to the handler with no
-descriptor-pool reflection, which keeps the Allocate hot path short.
+Transport: the in-house egrpc server (elastic_gpu_agent_amd/egrpc) — unary
+handlers dispatch inline on the connection thread with precomputed response
+header blocks, which keeps Allocate p50 ~6-10× below what grpcio's Python
+bindings achieve on the same socket (measured in tests/test_egrpc.py);
+wire-interop with the real (Go) kubelet stack is verified against grpcio in
+the test-suite. Registration to kubelet.sock uses the egrpc client.
 """
 from __future__ import annotations
 
@@ -18,12 +21,9 @@ import logging
 import os
 import threading
 import time
-from concurrent import futures
 from typing import Optional
 
-import grpc
-
-from .. import consts
+from .. import consts, egrpc
 from ..metrics import GLOBAL_METRICS
 from ..protos import deviceplugin as dp
 
@@ -41,93 +41,81 @@ class DevicePluginServer:
         plugin_dir: str = consts.DEVICE_PLUGIN_PATH,
         kubelet_socket: Optional[str] = None,
     ):
+        # (server instance is an egrpc.Server once served)
         self.plugin = plugin
         self.resource_name = resource_name
         self.endpoint = endpoint
         self.plugin_dir = plugin_dir
         self.kubelet_socket = kubelet_socket or os.path.join(plugin_dir, "kubelet.sock")
         self.socket_path = os.path.join(plugin_dir, endpoint)
-        self._server: Optional[grpc.Server] = None
+        self._server = None
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self._ready = threading.Event()
 
     # ---- gRPC service wiring ----
-    def _handlers(self):
+    def _methods(self):
         p = self.plugin
 
         def timed(name, fn, req_spec, resp_spec):
             metric = f"{self.resource_name}/{name}"
+            rec = GLOBAL_METRICS.recorder(metric)
+            decode = req_spec.decode if req_spec else None
+            encode = resp_spec.encode if resp_spec else None
 
             def handler(request_bytes, context):
-                with GLOBAL_METRICS.time(metric):
-                    req = req_spec.decode(request_bytes) if req_spec else {}
-                    resp = fn(req, context)
-                    return resp_spec.encode(resp)
+                t0 = time.perf_counter()
+                req = decode(request_bytes) if decode else {}
+                resp = fn(req, context)
+                out = encode(resp) if encode else b""
+                rec.observe(time.perf_counter() - t0)
+                return out
 
-            return grpc.unary_unary_rpc_method_handler(handler)
+            return egrpc.unary_unary(handler)
 
         def list_and_watch(request_bytes, context):
             for resp in p.list_and_watch(context):
                 yield dp.ListAndWatchResponse.encode(resp)
 
-        return grpc.method_handlers_generic_handler(
-            dp.DEVICE_PLUGIN_SERVICE,
-            {
-                "GetDevicePluginOptions": timed(
-                    "GetDevicePluginOptions",
-                    p.get_device_plugin_options,
-                    dp.Empty,
-                    dp.DevicePluginOptions,
-                ),
-                "ListAndWatch": grpc.unary_stream_rpc_method_handler(list_and_watch),
-                "GetPreferredAllocation": timed(
-                    "GetPreferredAllocation",
-                    p.get_preferred_allocation,
-                    dp.PreferredAllocationRequest,
-                    dp.PreferredAllocationResponse,
-                ),
-                "Allocate": timed(
-                    "Allocate", p.allocate, dp.AllocateRequest, dp.AllocateResponse
-                ),
-                "PreStartContainer": timed(
-                    "PreStartContainer",
-                    p.pre_start_container,
-                    dp.PreStartContainerRequest,
-                    dp.PreStartContainerResponse,
-                ),
-            },
-        )
+        return {
+            "GetDevicePluginOptions": timed(
+                "GetDevicePluginOptions", p.get_device_plugin_options,
+                dp.Empty, dp.DevicePluginOptions),
+            "ListAndWatch": egrpc.unary_stream(list_and_watch),
+            "GetPreferredAllocation": timed(
+                "GetPreferredAllocation", p.get_preferred_allocation,
+                dp.PreferredAllocationRequest, dp.PreferredAllocationResponse),
+            "Allocate": timed("Allocate", p.allocate, dp.AllocateRequest,
+                              dp.AllocateResponse),
+            "PreStartContainer": timed(
+                "PreStartContainer", p.pre_start_container,
+                dp.PreStartContainerRequest, dp.PreStartContainerResponse),
+        }
 
     # ---- lifecycle ----
     def serve(self) -> None:
-        if os.path.exists(self.socket_path):
-            os.unlink(self.socket_path)
         os.makedirs(self.plugin_dir, exist_ok=True)
-        self._server = grpc.server(
-            futures.ThreadPoolExecutor(max_workers=8),
-            options=[("grpc.max_receive_message_length", consts.POD_RESOURCES_MAX_SIZE)],
-        )
-        self._server.add_generic_rpc_handlers((self._handlers(),))
-        self._server.add_insecure_port(f"unix://{self.socket_path}")
+        self._server = egrpc.Server()
+        self._server.add_service(dp.DEVICE_PLUGIN_SERVICE, self._methods())
+        self._server.bind_unix(self.socket_path)
         self._server.start()
 
     def wait_ready(self, timeout: float = 5.0) -> None:
         """Self-dial the freshly served socket before registering
         (ref behavior: pkg/plugins/base.go:185-196)."""
-        deadline = time.time() + timeout
-        while time.time() < deadline:
-            try:
-                ch = grpc.insecure_channel(f"unix://{self.socket_path}")
-                grpc.channel_ready_future(ch).result(timeout=1.0)
-                ch.close()
-                return
-            except Exception:
-                time.sleep(0.05)
-        raise TimeoutError(f"plugin socket {self.socket_path} not ready")
+        ch = egrpc.Channel(self.socket_path, connect_timeout=timeout)
+        try:
+            get_opts = ch.unary_unary(
+                dp.METHOD_GET_OPTIONS,
+                request_serializer=dp.Empty.encode,
+                response_deserializer=dp.DevicePluginOptions.decode,
+            )
+            get_opts({}, timeout=timeout)
+        finally:
+            ch.close()
 
     def register(self) -> None:
-        ch = grpc.insecure_channel(f"unix://{self.kubelet_socket}")
+        ch = egrpc.Channel(self.kubelet_socket)
         try:
             register = ch.unary_unary(
                 dp.METHOD_REGISTER,

@@ -30,14 +30,6 @@ from .config import GPUPluginConfig
 log = logging.getLogger(__name__)
 
 
-def _rpc_error(context, code, msg):
-    if context is not None:
-        import grpc
-
-        context.abort(code, msg)
-    raise RuntimeError(msg)
-
-
 class GPUSharePluginBase:
     """Handlers shared by the core and memory resource servers."""
 
@@ -123,7 +115,10 @@ class GPUSharePluginBase:
         resp: dict = {"envs": {}, "mounts": []}
         if not (self.cfg.options.isolation and self.cfg.limits and paths.shim_host_path):
             return resp
-        limits_host = self.cfg.limits.touch(device.hash)
+        # NOTE: the limits file itself is written at PreStart (before the
+        # container is created); Allocate only declares the mount. Keeps the
+        # Allocate hot path free of disk I/O (p50 latency is the headline).
+        limits_host = self.cfg.limits.host_path(device.hash)
         resp["mounts"] = [
             {
                 "container_path": paths.shim_container_path,
@@ -189,9 +184,9 @@ class GPUSharePluginBase:
     def _fail(context, msg):
         log.error("%s", msg)
         if context is not None:
-            import grpc
+            from .. import egrpc
 
-            context.abort(grpc.StatusCode.INVALID_ARGUMENT, msg)
+            context.abort(egrpc.INVALID_ARGUMENT, msg)
         raise RuntimeError(msg)
 
 

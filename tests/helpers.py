@@ -84,10 +84,14 @@ class Harness:
 
 
 class PluginClient:
-    """gRPC client speaking the device-plugin API to a served socket."""
+    """egrpc client speaking the device-plugin API to a served socket.
+    (GrpcioPluginClient below drives the same socket with the real gRPC
+    stack for interop coverage.)"""
 
     def __init__(self, socket_path: str):
-        self.channel = grpc.insecure_channel(f"unix://{socket_path}")
+        from elastic_gpu_agent_amd import egrpc
+
+        self.channel = egrpc.Channel(socket_path)
         mk = self.channel.unary_unary
         self.get_options = mk(
             dp.METHOD_GET_OPTIONS,
@@ -117,6 +121,39 @@ class PluginClient:
 
     def close(self):
         self.channel.close()
+
+
+class GrpcioPluginClient(PluginClient):
+    """Same surface via grpcio — the 'real kubelet stack' direction."""
+
+    def __init__(self, socket_path: str):
+        self.channel = grpc.insecure_channel(f"unix://{socket_path}")
+        mk = self.channel.unary_unary
+        self.get_options = mk(
+            dp.METHOD_GET_OPTIONS,
+            request_serializer=dp.Empty.encode,
+            response_deserializer=dp.DevicePluginOptions.decode,
+        )
+        self.allocate = mk(
+            dp.METHOD_ALLOCATE,
+            request_serializer=dp.AllocateRequest.encode,
+            response_deserializer=dp.AllocateResponse.decode,
+        )
+        self.pre_start = mk(
+            dp.METHOD_PRE_START_CONTAINER,
+            request_serializer=dp.PreStartContainerRequest.encode,
+            response_deserializer=dp.PreStartContainerResponse.decode,
+        )
+        self.preferred = mk(
+            dp.METHOD_GET_PREFERRED_ALLOCATION,
+            request_serializer=dp.PreferredAllocationRequest.encode,
+            response_deserializer=dp.PreferredAllocationResponse.decode,
+        )
+        self.list_and_watch = self.channel.unary_stream(
+            dp.METHOD_LIST_AND_WATCH,
+            request_serializer=dp.Empty.encode,
+            response_deserializer=dp.ListAndWatchResponse.decode,
+        )
 
 
 class FakeKubeletRegistration:

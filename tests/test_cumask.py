@@ -15,8 +15,8 @@ def test_percent_to_cu_count():
     assert cu_count_for_percent(150) == 256  # clamped to a full card
     assert cu_count_for_percent(50) == 128
     assert cu_count_for_percent(25) == 64
-    assert cu_count_for_percent(1) == 3  # 2.56 rounds to 3
-    assert cu_count_for_percent(0) == 1  # never zero CUs
+    assert cu_count_for_percent(1) == 4  # 2.56 → 3 → rounded up to a CU pair
+    assert cu_count_for_percent(0) == 2  # never less than one CU pair
 
 
 def test_round_robin_spreads_across_xcds():
@@ -29,12 +29,19 @@ def test_round_robin_spreads_across_xcds():
 
 
 def test_round_robin_uneven_spread():
-    cus = xcd_round_robin_cus(13)
+    cus = xcd_round_robin_cus(13)  # 13 CUs → 7 pairs
     per_xcd = [0] * 8
     for cu in cus:
         per_xcd[cu // 32] += 1
-    # 13 = 8*1 + 5 extras on the first five XCDs
-    assert per_xcd == [2, 2, 2, 2, 2, 1, 1, 1]
+    # one pair on each of the first seven XCDs
+    assert per_xcd == [2, 2, 2, 2, 2, 2, 2, 0]
+
+
+def test_pair_granularity():
+    cus = xcd_round_robin_cus(64)
+    s = set(cus)
+    for cu in s:
+        assert cu ^ 1 in s  # every CU's pair sibling is present
 
 
 def test_offset_rotates_within_xcd():

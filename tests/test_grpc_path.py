@@ -47,7 +47,7 @@ def test_serve_register_and_rpcs(h, tmp_path):
             stream = client.list_and_watch({})
             first = next(stream)
             assert len(first["devices"]) == 200
-            stream.cancel()
+            stream.close()  # generator: releases the channel for unary calls
 
             # Allocate + PreStart over the wire
             ids = [f"1-{i:02d}" for i in range(40)]
@@ -62,12 +62,38 @@ def test_serve_register_and_rpcs(h, tmp_path):
             ) == "/dev/dri/renderD129"
 
             # error path surfaces as INVALID_ARGUMENT
+            from elastic_gpu_agent_amd import egrpc
+
             bad = [f"0-{i:02d}" for i in range(10)]
-            with pytest.raises(grpc.RpcError) as ei:
+            with pytest.raises(egrpc.EgrpcError) as ei:
                 client.pre_start({"devicesIDs": bad})
-            assert ei.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+            assert ei.value.code() == egrpc.INVALID_ARGUMENT
         finally:
             client.close()
+
+        # ---- the same socket driven by the REAL gRPC stack (kubelet dir.) --
+        from helpers import GrpcioPluginClient
+
+        gclient = GrpcioPluginClient(h.plugin.core_server.socket_path)
+        try:
+            opts = gclient.get_options({})
+            assert opts["pre_start_required"] is True
+            stream = gclient.list_and_watch({})
+            first = next(stream)
+            assert len(first["devices"]) == 200
+            stream.cancel()
+            ids2 = [f"0-{i:02d}" for i in range(60, 80)]
+            d2 = Device.new(ids2, consts.RESOURCE_GPU_CORE)
+            h.core_locator.assign(d2.hash, PodContainer("ns", "gp2", "main"))
+            h.add_assumed_pod("ns", "gp2", "main", "0")
+            resp = gclient.allocate({"container_requests": [{"devicesIDs": ids2}]})
+            assert resp["container_responses"][0]["envs"]["GPU"] == d2.hash
+            gclient.pre_start({"devicesIDs": ids2})
+            with pytest.raises(grpc.RpcError) as gei:
+                gclient.pre_start({"devicesIDs": ["9-99"]})
+            assert gei.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+        finally:
+            gclient.close()
     finally:
         kubelet.stop()
 

@@ -53,8 +53,9 @@ def test_core_allocate_fractional(h):
     paths = [s["host_path"] for s in cr["devices"]]
     assert consts.KFD_PATH in paths
     assert f"/dev/elastic-gpu-{d.hash}-0" in paths
-    # limits file exists (touched) so kubelet can mount it
-    assert os.path.exists(h.plugin.cfg.limits.host_path(d.hash))
+    # limits file is only declared here; it is written at PreStart (keeps
+    # Allocate free of disk I/O)
+    assert not os.path.exists(h.plugin.cfg.limits.host_path(d.hash))
 
 
 def test_core_allocate_whole_gpus_no_shim(h):

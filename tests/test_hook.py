@@ -1,0 +1,94 @@
+"""OCI prestart hook tests (dry-run mode; no namespaces touched).
+
+The binary (native/egpu_hook.cpp) replaces the reference's Go hook + its
+prebuilt forked nvidia toolkit (SURVEY #23/#24): state JSON on stdin →
+GPU=<hash> env from the bundle config → elastic-gpu links → device nodes.
+"""
+import json
+import os
+import subprocess
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+HOOK = os.path.join(REPO, "bin", "egpu-hook")
+
+
+@pytest.fixture(scope="module", autouse=True)
+def build_hook():
+    if not os.path.exists(HOOK):
+        subprocess.check_call(
+            ["python", "-m", "elastic_gpu_agent_amd.native.build"], cwd=REPO
+        )
+
+
+def run_hook(state: dict, dev_root: str, log_path: str, arg="prestart"):
+    env = dict(os.environ)
+    env["EGPU_HOOK_DRYRUN"] = "1"
+    env["EGPU_DEV_ROOT"] = dev_root
+    env["EGPU_HOOK_LOG"] = log_path
+    return subprocess.run(
+        [HOOK, arg], input=json.dumps(state).encode(), env=env,
+        capture_output=True, timeout=30,
+    )
+
+
+def make_bundle(tmp_path, envs):
+    bundle = tmp_path / "bundle"
+    bundle.mkdir(exist_ok=True)
+    config = {"ociVersion": "1.0.2", "process": {"env": envs, "args": ["sleep"]}}
+    (bundle / "config.json").write_text(json.dumps(config))
+    return str(bundle)
+
+
+def test_hook_injects_for_gpu_env(tmp_path):
+    dev = tmp_path / "dev"
+    (dev / "dri").mkdir(parents=True)
+    os.symlink("/dev/dri/renderD129", str(dev / "elastic-gpu-cafe1234-0"))
+    os.symlink("/dev/kfd", str(dev / "elastic-gpuctl-cafe1234-0"))
+    bundle = make_bundle(tmp_path, ["PATH=/bin", "GPU=cafe1234"])
+    state = {"ociVersion": "1.0.2", "id": "c1", "pid": os.getpid(), "bundle": bundle}
+    r = run_hook(state, str(dev), str(tmp_path / "hook.log"))
+    assert r.returncode == 0, r.stderr
+    plan = json.loads(r.stdout)
+    paths = {n["path"]: (n["major"], n["minor"]) for n in plan["nodes"]}
+    assert "/dev/kfd" in paths
+    assert paths["/dev/dri/renderD129"] == (226, 129)  # minor from link target
+
+
+def test_hook_passthrough_without_gpu_env(tmp_path):
+    bundle = make_bundle(tmp_path, ["PATH=/bin"])
+    state = {"ociVersion": "1.0.2", "id": "c2", "pid": 1, "bundle": bundle}
+    r = run_hook(state, str(tmp_path), str(tmp_path / "hook.log"))
+    assert r.returncode == 0
+    assert r.stdout.strip() == b""  # no injection plan
+
+
+def test_hook_fails_when_links_missing(tmp_path):
+    dev = tmp_path / "dev"
+    dev.mkdir()
+    bundle = make_bundle(tmp_path, ["GPU=deadbeef"])
+    state = {"ociVersion": "1.0.2", "id": "c3", "pid": 1, "bundle": bundle}
+    r = run_hook(state, str(dev), str(tmp_path / "hook.log"))
+    assert r.returncode != 0
+    log = (tmp_path / "hook.log").read_text()
+    assert "deadbeef" in log
+
+
+def test_hook_multi_gpu_links(tmp_path):
+    dev = tmp_path / "dev"
+    dev.mkdir()
+    os.symlink("/dev/dri/renderD128", str(dev / "elastic-gpu-aa11bb22-0"))
+    os.symlink("/dev/dri/renderD130", str(dev / "elastic-gpu-aa11bb22-1"))
+    bundle = make_bundle(tmp_path, ["GPU=aa11bb22"])
+    state = {"ociVersion": "1.0.2", "id": "c4", "pid": 1, "bundle": bundle}
+    r = run_hook(state, str(dev), str(tmp_path / "hook.log"))
+    assert r.returncode == 0, r.stderr
+    plan = json.loads(r.stdout)
+    minors = sorted(n["minor"] for n in plan["nodes"] if n["path"].startswith("/dev/dri"))
+    assert minors == [128, 130]
+
+
+def test_hook_ignores_other_lifecycle_args(tmp_path):
+    r = subprocess.run([HOOK, "poststop"], input=b"{}", capture_output=True, timeout=30)
+    assert r.returncode == 0

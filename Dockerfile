@@ -1,0 +1,18 @@
+# Agent image for the MI355X-native elastic-gpu agent.
+# Build on a ROCm base so hipcc can compile the gfx950 kernels and the
+# HSA shim at image build time (no JIT cache at runtime).
+FROM rocm/dev-ubuntu-22.04:7.0 AS build
+WORKDIR /src
+COPY elastic_gpu_agent_amd/ elastic_gpu_agent_amd/
+RUN python3 -m pip install --no-cache-dir pybind11 && \
+    python3 -m elastic_gpu_agent_amd.native.build
+
+FROM rocm/dev-ubuntu-22.04:7.0
+RUN python3 -m pip install --no-cache-dir grpcio httpx pyyaml prometheus_client
+WORKDIR /opt/agent
+COPY --from=build /src/elastic_gpu_agent_amd /opt/agent/elastic_gpu_agent_amd
+COPY --from=build /src/bin/egpu-hook /opt/egpu/egpu-hook
+COPY tools/install.sh /opt/egpu/install.sh
+COPY bench.py /opt/agent/bench.py
+ENV PYTHONPATH=/opt/agent
+ENTRYPOINT ["python3", "-m", "elastic_gpu_agent_amd.cli.agent"]

@@ -46,13 +46,15 @@ __global__ void __launch_bounds__(256) census_kernel(uint32_t* out, int spin) {
   asm volatile("s_getreg_b32 %0, hwreg(HW_REG_HW_ID)" : "=s"(hwid));
   asm volatile("s_getreg_b32 %0, hwreg(HW_REG_XCC_ID)" : "=s"(xcc));
   // keep the block resident for a while so co-resident blocks spread over
-  // every enabled CU instead of a few fast-retiring ones
-  float acc = threadIdx.x * 1.0f;
+  // every enabled CU instead of a few fast-retiring ones; the accumulator's
+  // sign bit feeds the result so the loop cannot be dead-code-eliminated
+  // (acc stays positive, so the recorded value is unchanged in practice)
+  float acc = threadIdx.x * 1.0f + 0.25f;
   for (int i = 0; i < spin; ++i) acc = fmaf(acc, 1.0000001f, 0.5f);
   if (threadIdx.x == 0) {
     // bits [14:8] of HW_ID = {SE_ID[14:13], SH_ID[12], CU_ID[11:8]} on gfx9
     // lineage; XCC_ID identifies the XCD. Together: one physical CU.
-    out[blockIdx.x] = (xcc << 16) | ((hwid >> 8) & 0x7F) | ((acc > 1e30f) ? 0 : 0);
+    out[blockIdx.x] = (xcc << 16) | ((hwid >> 8) & 0x7F) | (__float_as_uint(acc) >> 31 << 30);
   }
 }
 

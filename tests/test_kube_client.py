@@ -1,0 +1,202 @@
+"""K8sClient + PodSitter tests against a stub Kubernetes API server
+(plain HTTP server speaking the pods list/watch/get subset)."""
+import json
+import threading
+import time
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from urllib.parse import parse_qs, urlparse
+
+import pytest
+
+from elastic_gpu_agent_amd.kube.client import K8sClient, NotFound
+from elastic_gpu_agent_amd.kube.sitter import PodSitter
+
+
+class StubK8s:
+    """Minimal pods API: list with fieldSelector, watch (chunked), get."""
+
+    def __init__(self):
+        self.pods = {}  # (ns, name) -> pod dict
+        self.rv = 1
+        self.watch_events = []  # queued events served to the next watch
+        self._watch_cv = threading.Condition()
+        stub = self
+
+        class Handler(BaseHTTPRequestHandler):
+            protocol_version = "HTTP/1.1"
+
+            def log_message(self, *a):
+                pass
+
+            def do_GET(self):
+                u = urlparse(self.path)
+                q = parse_qs(u.query)
+                parts = u.path.strip("/").split("/")
+                if u.path == "/api/v1/pods" and q.get("watch", ["false"])[0] != "true":
+                    node = ""
+                    fs = q.get("fieldSelector", [""])[0]
+                    if fs.startswith("spec.nodeName="):
+                        node = fs.split("=", 1)[1]
+                    items = [p for p in stub.pods.values()
+                             if not node or p["spec"].get("nodeName") == node]
+                    body = json.dumps({
+                        "kind": "PodList",
+                        "metadata": {"resourceVersion": str(stub.rv)},
+                        "items": items,
+                    }).encode()
+                    self.send_response(200)
+                    self.send_header("Content-Type", "application/json")
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                elif u.path == "/api/v1/pods":  # watch
+                    self.send_response(200)
+                    self.send_header("Content-Type", "application/json")
+                    self.send_header("Transfer-Encoding", "chunked")
+                    self.end_headers()
+                    deadline = time.time() + 5.0
+                    sent = 0
+                    while time.time() < deadline:
+                        with stub._watch_cv:
+                            if sent < len(stub.watch_events):
+                                ev = stub.watch_events[sent]
+                                sent += 1
+                            else:
+                                stub._watch_cv.wait(timeout=0.2)
+                                continue
+                        line = (json.dumps(ev) + "\n").encode()
+                        try:
+                            self.wfile.write(f"{len(line):x}\r\n".encode() + line + b"\r\n")
+                            self.wfile.flush()
+                        except OSError:
+                            return
+                    try:
+                        self.wfile.write(b"0\r\n\r\n")
+                    except OSError:
+                        pass
+                elif len(parts) == 6 and parts[2] == "namespaces" and parts[4] == "pods":
+                    ns, name = parts[3], parts[5]
+                    pod = stub.pods.get((ns, name))
+                    if pod is None:
+                        body = json.dumps({"kind": "Status", "code": 404}).encode()
+                        self.send_response(404)
+                    else:
+                        body = json.dumps(pod).encode()
+                        self.send_response(200)
+                    self.send_header("Content-Type", "application/json")
+                    self.send_header("Content-Length", str(len(body)))
+                    self.end_headers()
+                    self.wfile.write(body)
+                else:
+                    self.send_response(404)
+                    self.send_header("Content-Length", "0")
+                    self.end_headers()
+
+        self.server = ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+        self.port = self.server.server_address[1]
+        threading.Thread(target=self.server.serve_forever, daemon=True).start()
+
+    def add_pod(self, ns, name, node="n1", annotations=None):
+        pod = {
+            "metadata": {"namespace": ns, "name": name, "uid": f"uid-{name}",
+                         "annotations": annotations or {}},
+            "spec": {"nodeName": node},
+            "status": {"phase": "Running"},
+        }
+        self.pods[(ns, name)] = pod
+        return pod
+
+    def push_event(self, etype, pod):
+        with self._watch_cv:
+            self.watch_events.append({"type": etype, "object": pod})
+            self._watch_cv.notify_all()
+
+    def stop(self):
+        self.server.shutdown()
+
+
+@pytest.fixture
+def stub():
+    s = StubK8s()
+    yield s
+    s.stop()
+
+
+def test_list_pods_filters_by_node(stub):
+    stub.add_pod("ns", "a", node="n1")
+    stub.add_pod("ns", "b", node="n2")
+    c = K8sClient(base_url=f"http://127.0.0.1:{stub.port}")
+    pods, rv = c.list_pods("n1")
+    assert [p.name for p in pods] == ["a"]
+    assert rv == str(stub.rv)
+    c.close()
+
+
+def test_get_pod_and_notfound(stub):
+    stub.add_pod("ns", "a", annotations={"elasticgpu.io/assumed": "true"})
+    c = K8sClient(base_url=f"http://127.0.0.1:{stub.port}")
+    p = c.get_pod("ns", "a")
+    assert p.is_assumed()
+    with pytest.raises(NotFound):
+        c.get_pod("ns", "missing")
+    c.close()
+
+
+def test_sitter_sync_watch_and_delete_hook(stub):
+    pod = stub.add_pod("ns", "w1", node="n1")
+    deleted = []
+    c = K8sClient(base_url=f"http://127.0.0.1:{stub.port}")
+    sitter = PodSitter(c, "n1", delete_hook=lambda p: deleted.append(p.name))
+    sitter.start()
+    deadline = time.time() + 5
+    while not sitter.has_synced() and time.time() < deadline:
+        time.sleep(0.05)
+    assert sitter.has_synced()
+    assert sitter.get_pod("ns", "w1").name == "w1"
+
+    # ADDED via watch
+    pod2 = stub.add_pod("ns", "w2", node="n1")
+    stub.push_event("ADDED", pod2)
+    deadline = time.time() + 5
+    while time.time() < deadline:
+        try:
+            sitter.get_pod("ns", "w2")
+            break
+        except Exception:
+            time.sleep(0.05)
+    assert sitter.get_pod("ns", "w2").name == "w2"
+
+    # DELETED via watch fires the GC hook
+    stub.push_event("DELETED", pod)
+    deadline = time.time() + 5
+    while "w1" not in deleted and time.time() < deadline:
+        time.sleep(0.05)
+    assert "w1" in deleted
+    with pytest.raises(Exception):
+        sitter.get_pod("ns", "w1")
+    sitter.stop()
+    c.close()
+
+
+def test_kubeconfig_parsing(tmp_path):
+    kc = tmp_path / "kubeconfig"
+    kc.write_text(
+        """
+apiVersion: v1
+kind: Config
+current-context: ctx
+contexts:
+- name: ctx
+  context: {cluster: c1, user: u1}
+clusters:
+- name: c1
+  cluster: {server: "http://127.0.0.1:1", insecure-skip-tls-verify: true}
+users:
+- name: u1
+  user: {token: sekrit}
+"""
+    )
+    c = K8sClient(kubeconf=str(kc))
+    assert c._client.headers["Authorization"] == "Bearer sekrit"
+    assert str(c._client.base_url).startswith("http://127.0.0.1:1")
+    c.close()

@@ -1,0 +1,28 @@
+#!/bin/sh
+# Host installer run by the DaemonSet init container (privileged, with
+# /host/usr/local/bin, /host/opt/egpu and /host/etc/containers/oci/hooks.d
+# mounted). MI355X-native replacement for the reference's hook swap
+# (ref: tools/install.sh replaced nvidia-container-runtime-hook): instead of
+# hijacking a vendor hook we install our own first-class one.
+set -e
+
+# 1. the OCI prestart hook binary
+cp /opt/egpu/egpu-hook /host/usr/local/bin/egpu-hook
+chmod 0755 /host/usr/local/bin/egpu-hook
+
+# 2. the HSA shim the agent mounts into fractional pods
+mkdir -p /host/opt/egpu
+cp /opt/agent/elastic_gpu_agent_amd/libegpu_shim.so /host/opt/egpu/libegpu_shim.so
+
+# 3. OCI hooks.d registration (CRI-O / podman style). containerd users add
+#    the hook via the runtime handler config instead; see docs/DEPLOY.md.
+mkdir -p /host/etc/containers/oci/hooks.d
+cat > /host/etc/containers/oci/hooks.d/10-egpu.json <<'EOF'
+{
+  "version": "1.0.0",
+  "hook": {"path": "/usr/local/bin/egpu-hook", "args": ["egpu-hook", "prestart"]},
+  "when": {"hasBindMounts": false, "annotations": {".*": ".*"}},
+  "stages": ["prestart"]
+}
+EOF
+echo "egpu host components installed"

@@ -1,0 +1,109 @@
+"""egpuctl — node-local admin CLI for the MI355X elastic-gpu agent.
+
+Subcommands:
+  devices                 enumerate GPUs (amdsmi or --fake)
+  pods                    list persisted allocations
+  masks                   list live CU-mask assignments
+  occupancy               per-pod live occupancy (amdsmi join)
+  gc                      run one GC reconciliation pass (needs cluster access)
+  migrate --from BOLT_DB  import a reference agent's BoltDB state
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+
+def cmd_devices(args) -> int:
+    if args.fake:
+        from ..operator.fake import FakeBackend
+
+        devices = FakeBackend().devices()
+    else:
+        from ..operator.amdsmi import AmdSmiBackend
+
+        devices = AmdSmiBackend().devices()
+    import dataclasses
+
+    print(json.dumps([dataclasses.asdict(d) for d in devices], indent=2))
+    return 0
+
+
+def cmd_pods(args) -> int:
+    from ..storage import Storage
+
+    st = Storage(args.db)
+    rows = []
+    st.for_each(
+        lambda pi: rows.append(
+            {
+                "pod": pi.key(),
+                "containers": {
+                    c: {"hash": d.hash, "resource": d.resource_name, "units": len(d.list)}
+                    for c, d in pi.container_device_map.items()
+                },
+            }
+        )
+    )
+    print(json.dumps(rows, indent=2))
+    st.close()
+    return 0
+
+
+def cmd_masks(args) -> int:
+    from ..isolation import AUX_MASK_PREFIX
+    from ..storage import Storage
+
+    st = Storage(args.db)
+    out = {k[len(AUX_MASK_PREFIX):]: json.loads(v) for k, v in st.aux_items(AUX_MASK_PREFIX)}
+    print(json.dumps(out, indent=2))
+    st.close()
+    return 0
+
+
+def cmd_occupancy(args) -> int:
+    from ..isolation import LimitsWriter
+    from ..isolation.occupancy import report
+    from ..storage import Storage
+
+    st = Storage(args.db)
+    limits = LimitsWriter(args.limits_dir)
+    print(json.dumps(report(st, limits, state_dir=args.state_dir), indent=2))
+    st.close()
+    return 0
+
+
+def cmd_migrate(args) -> int:
+    from ..storage import Storage, migrate_from_bolt
+
+    st = Storage(args.db)
+    n = migrate_from_bolt(getattr(args, "from"), st)
+    print(f"migrated {n} pod records")
+    st.close()
+    return 0
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="egpuctl")
+    p.add_argument("--db", default="/host/var/lib/egpu/meta.db")
+    p.add_argument("--limits-dir", default="/host/var/lib/egpu/limits")
+    p.add_argument("--state-dir", default="/host/var/lib/egpu")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    d = sub.add_parser("devices")
+    d.add_argument("--fake", action="store_true")
+    d.set_defaults(fn=cmd_devices)
+    sub.add_parser("pods").set_defaults(fn=cmd_pods)
+    sub.add_parser("masks").set_defaults(fn=cmd_masks)
+    sub.add_parser("occupancy").set_defaults(fn=cmd_occupancy)
+    m = sub.add_parser("migrate")
+    m.add_argument("--from", required=True)
+    m.set_defaults(fn=cmd_migrate)
+
+    args = p.parse_args(argv)
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

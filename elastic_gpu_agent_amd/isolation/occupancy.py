@@ -1,0 +1,102 @@
+"""Per-pod GPU occupancy reporting.
+
+Joins three sources into one view (the observability the reference lacked —
+SURVEY §5 'per-pod occupancy via rocprof/amdsmi counters'):
+
+- the agent's persisted allocations (storage: pod → container → device hash),
+- the isolation metadata (aux table: CU mask / limits per hash; limits files),
+- live amdsmi telemetry: per-GPU engine busy% + VRAM, and the per-process
+  list (vram_bytes, cu_occupancy — CUs the process occupies right now),
+  attributed to pods via the hash→pid files the OCI hook records.
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Dict, List, Optional
+
+from ..types import PodInfo
+from . import AUX_MASK_PREFIX
+
+
+def _read_pid(state_dir: str, alloc_hash: str) -> Optional[int]:
+    try:
+        with open(os.path.join(state_dir, "pids", alloc_hash)) as f:
+            return int(f.read().strip())
+    except (OSError, ValueError):
+        return None
+
+
+def report(
+    storage,
+    limits_writer=None,
+    state_dir: str = "/var/lib/egpu",
+    smi=None,
+) -> dict:
+    """Returns {"gpus": {index: telemetry}, "pods": [per-allocation rows]}."""
+    if smi is None:
+        from elastic_gpu_agent_amd import _amdsmi as smi  # native, no fallback
+
+    pods: List[PodInfo] = []
+    storage.for_each(pods.append)
+
+    # mask metadata per hash
+    masks: Dict[str, dict] = {}
+    for key, val in storage.aux_items(AUX_MASK_PREFIX):
+        masks[key[len(AUX_MASK_PREFIX):]] = json.loads(val)
+
+    gpu_indexes = set()
+    rows = []
+    for pi in pods:
+        for container, device in pi.container_device_map.items():
+            rec = masks.get(device.hash, {})
+            gpu_index = rec.get("gpu_index")
+            limits = {}
+            if limits_writer is not None:
+                try:
+                    limits = limits_writer.read(device.hash)
+                except (OSError, ValueError):
+                    limits = {}
+            if gpu_index is None:
+                gis = limits.get("gpu_indexes") or []
+                gpu_index = gis[0] if gis else None
+            if gpu_index is not None:
+                gpu_indexes.add(gpu_index)
+            rows.append({
+                "pod": pi.key(),
+                "container": container,
+                "hash": device.hash,
+                "resource": device.resource_name,
+                "units": len(device.list),
+                "gpu_index": gpu_index,
+                "cu_limit": rec.get("cu_count") or limits.get("cu_count"),
+                "mem_limit_bytes": limits.get("mem_limit_bytes"),
+                "pid": _read_pid(state_dir, device.hash),
+            })
+
+    gpus: Dict[int, dict] = {}
+    procs_by_gpu: Dict[int, list] = {}
+    for idx in sorted(gpu_indexes):
+        try:
+            gpus[idx] = dict(smi.gpu_utilization(idx))
+            procs_by_gpu[idx] = list(smi.gpu_processes(idx))
+        except Exception as e:  # telemetry is best-effort; allocations are not
+            gpus[idx] = {"error": str(e)}
+            procs_by_gpu[idx] = []
+
+    # attribute live process telemetry to pods via recorded pids
+    for row in rows:
+        pid, idx = row["pid"], row["gpu_index"]
+        row["live"] = None
+        if pid is None or idx not in procs_by_gpu:
+            continue
+        for proc in procs_by_gpu[idx]:
+            if proc.get("pid") == pid:
+                row["live"] = {
+                    "vram_bytes": proc.get("vram_bytes"),
+                    "cu_occupancy": proc.get("cu_occupancy"),
+                    "gfx_busy_ns": proc.get("gfx_busy_ns"),
+                }
+                break
+
+    return {"gpus": gpus, "pods": rows}

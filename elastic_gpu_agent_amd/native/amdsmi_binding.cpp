@@ -157,10 +157,69 @@ py::list enumerate_gpus() {
   return out;
 }
 
+// ---- occupancy / utilization (per-pod occupancy reporting) -----------------
+
+amdsmi_processor_handle handle_for_hip_index(uint32_t index,
+                                             std::vector<amdsmi_processor_handle>& gpus) {
+  for (auto h : gpus) {
+    amdsmi_enumeration_info_t en{};
+    if (amdsmi_get_gpu_enumeration_info(h, &en) == AMDSMI_STATUS_SUCCESS &&
+        en.hip_id == index)
+      return h;
+  }
+  throw std::runtime_error("no GPU with HIP index " + std::to_string(index));
+}
+
+py::dict gpu_utilization(uint32_t index) {
+  SmiSession session;
+  auto gpus = gpu_handles();
+  auto h = handle_for_hip_index(index, gpus);
+  py::dict d;
+  amdsmi_engine_usage_t usage{};
+  if (amdsmi_get_gpu_activity(h, &usage) == AMDSMI_STATUS_SUCCESS) {
+    d["gfx_busy_percent"] = usage.gfx_activity;
+    d["mem_busy_percent"] = usage.umc_activity;
+  }
+  amdsmi_vram_usage_t vram{};
+  if (amdsmi_get_gpu_vram_usage(h, &vram) == AMDSMI_STATUS_SUCCESS) {
+    d["vram_used_mb"] = vram.vram_used;
+    d["vram_total_mb"] = vram.vram_total;
+  }
+  return d;
+}
+
+py::list gpu_processes(uint32_t index) {
+  SmiSession session;
+  auto gpus = gpu_handles();
+  auto h = handle_for_hip_index(index, gpus);
+  uint32_t n = 0;
+  amdsmi_status_t st = amdsmi_get_gpu_process_list(h, &n, nullptr);
+  py::list out;
+  if (st != AMDSMI_STATUS_SUCCESS || n == 0) return out;
+  std::vector<amdsmi_proc_info_t> procs(n);
+  st = amdsmi_get_gpu_process_list(h, &n, procs.data());
+  if (st != AMDSMI_STATUS_SUCCESS) return out;
+  for (uint32_t i = 0; i < n; ++i) {
+    py::dict p;
+    p["pid"] = static_cast<uint64_t>(procs[i].pid);
+    p["name"] = std::string(procs[i].name);
+    p["vram_bytes"] = procs[i].memory_usage.vram_mem;
+    p["gfx_busy_ns"] = procs[i].engine_usage.gfx;
+    p["cu_occupancy"] = procs[i].cu_occupancy;  // CUs this process occupies NOW
+    p["evicted_ms"] = procs[i].evicted_time;
+    out.append(p);
+  }
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_amdsmi, m) {
-  m.doc() = "MI355X enumeration via libamd_smi";
+  m.doc() = "MI355X enumeration + occupancy via libamd_smi";
   m.def("enumerate_gpus", &enumerate_gpus,
         "Enumerate AMD GPUs; returns list of dicts sorted by HIP id");
+  m.def("gpu_utilization", &gpu_utilization,
+        "Engine busy % + VRAM usage for one GPU (by HIP index)");
+  m.def("gpu_processes", &gpu_processes,
+        "Per-process VRAM / CU-occupancy list for one GPU (by HIP index)");
 }

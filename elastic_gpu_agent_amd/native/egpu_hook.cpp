@@ -335,5 +335,20 @@ int main(int argc, char** argv) {
 
   logf("injecting %zu nodes for GPU=%s into pid %ld%s", nodes.size(), hash.c_str(), pid,
        dryrun ? " (dry-run)" : "");
-  return inject(pid, nodes, dryrun);
+  int rc = inject(pid, nodes, dryrun);
+  if (rc == 0) {
+    // record hash→pid so the agent can attribute per-process GPU occupancy
+    // (amdsmi process list) back to pods
+    const char* state_env = getenv("EGPU_STATE_DIR");
+    std::string state_dir = state_env ? state_env : "/var/lib/egpu";
+    std::string pid_dir = state_dir + "/pids";
+    mkdir(state_dir.c_str(), 0755);
+    mkdir(pid_dir.c_str(), 0755);
+    FILE* pf = fopen((pid_dir + "/" + hash).c_str(), "w");
+    if (pf) {
+      fprintf(pf, "%ld\n", pid);
+      fclose(pf);
+    }
+  }
+  return rc;
 }

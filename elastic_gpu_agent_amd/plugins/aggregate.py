@@ -90,6 +90,14 @@ class GPUSharePlugin:
                     self.cfg.cumask.release(device.hash)
                 if self.cfg.limits:
                     self.cfg.limits.delete(device.hash)
+                try:  # hook-recorded pid file (occupancy attribution)
+                    import os
+
+                    os.unlink(
+                        os.path.join(self.cfg.paths.state_dir, "pids", device.hash)
+                    )
+                except OSError:
+                    pass
             self.cfg.storage.delete(pi.namespace, pi.name)
             log.info("GC reclaimed %s/%s", pi.namespace, pi.name)
         return len(doomed)

@@ -17,6 +17,7 @@ class AgentPaths:
     kubelet_socket: Optional[str] = None  # default: <plugin_dir>/kubelet.sock
     podresources_socket: str = consts.POD_RESOURCES_SOCKET
     limits_dir: str = "/host/var/lib/egpu/limits"
+    state_dir: str = "/host/var/lib/egpu"  # hook-recorded pids live here
     # host path of the HSA shim library mounted into containers; None disables
     # isolation env injection entirely.
     shim_host_path: Optional[str] = "/host/opt/egpu/libegpu_shim.so"

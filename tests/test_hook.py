@@ -22,11 +22,12 @@ def build_hook():
         )
 
 
-def run_hook(state: dict, dev_root: str, log_path: str, arg="prestart"):
+def run_hook(state: dict, dev_root: str, log_path: str, arg="prestart", state_dir=None):
     env = dict(os.environ)
     env["EGPU_HOOK_DRYRUN"] = "1"
     env["EGPU_DEV_ROOT"] = dev_root
     env["EGPU_HOOK_LOG"] = log_path
+    env["EGPU_STATE_DIR"] = state_dir or os.path.join(os.path.dirname(log_path), "state")
     return subprocess.run(
         [HOOK, arg], input=json.dumps(state).encode(), env=env,
         capture_output=True, timeout=30,
@@ -54,6 +55,9 @@ def test_hook_injects_for_gpu_env(tmp_path):
     paths = {n["path"]: (n["major"], n["minor"]) for n in plan["nodes"]}
     assert "/dev/kfd" in paths
     assert paths["/dev/dri/renderD129"] == (226, 129)  # minor from link target
+    # hash→pid recorded for occupancy attribution
+    pid_file = tmp_path / "state" / "pids" / "cafe1234"
+    assert pid_file.read_text().strip() == str(os.getpid())
 
 
 def test_hook_passthrough_without_gpu_env(tmp_path):

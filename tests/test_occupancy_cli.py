@@ -1,0 +1,99 @@
+"""Occupancy report + egpuctl CLI tests (fake smi module on CPU)."""
+import json
+import os
+
+import pytest
+
+from elastic_gpu_agent_amd import consts
+from elastic_gpu_agent_amd.isolation import LimitsWriter
+from elastic_gpu_agent_amd.isolation.occupancy import report
+from elastic_gpu_agent_amd.storage import Storage
+from elastic_gpu_agent_amd.types import Device, PodInfo
+
+
+class FakeSmi:
+    def gpu_utilization(self, idx):
+        return {"gfx_busy_percent": 42, "vram_used_mb": 1024, "vram_total_mb": 294912}
+
+    def gpu_processes(self, idx):
+        return [{"pid": 4242, "name": "python", "vram_bytes": 1 << 30,
+                 "cu_occupancy": 61, "gfx_busy_ns": 123456, "evicted_ms": 0}]
+
+
+@pytest.fixture
+def populated(tmp_path):
+    st = Storage(str(tmp_path / "meta.db"))
+    limits = LimitsWriter(str(tmp_path / "limits"))
+    pi = PodInfo("ns", "p1")
+    d = Device.new([f"0-{i:02d}" for i in range(25)], consts.RESOURCE_GPU_CORE)
+    pi.container_device_map["main"] = d
+    st.save(pi)
+    st.aux_set("mask/" + d.hash, json.dumps(
+        {"gpu_index": 0, "cu_mask": "03030303", "cu_count": 64, "percent": 25}))
+    limits.finalize(d.hash, [0], [], cu_mask="03030303", cu_count=64)
+    # hook-recorded pid
+    os.makedirs(tmp_path / "pids")
+    (tmp_path / "pids" / d.hash).write_text("4242\n")
+    yield st, limits, str(tmp_path), d
+    st.close()
+
+
+def test_occupancy_report_joins_sources(populated):
+    st, limits, state_dir, d = populated
+    rep = report(st, limits, state_dir=state_dir, smi=FakeSmi())
+    assert rep["gpus"][0]["gfx_busy_percent"] == 42
+    row = rep["pods"][0]
+    assert row["pod"] == "ns/p1" and row["hash"] == d.hash
+    assert row["cu_limit"] == 64
+    assert row["pid"] == 4242
+    assert row["live"]["cu_occupancy"] == 61
+    assert row["live"]["vram_bytes"] == 1 << 30
+
+
+def test_occupancy_without_pid(populated):
+    st, limits, state_dir, d = populated
+    os.unlink(os.path.join(state_dir, "pids", d.hash))
+    rep = report(st, limits, state_dir=state_dir, smi=FakeSmi())
+    assert rep["pods"][0]["pid"] is None
+    assert rep["pods"][0]["live"] is None
+
+
+def test_egpuctl_pods_and_masks(populated, capsys):
+    st, limits, state_dir, d = populated
+    from elastic_gpu_agent_amd.cli.egpuctl import main
+
+    db = os.path.join(state_dir, "meta.db")
+    assert main(["--db", db, "pods"]) == 0
+    out = json.loads(capsys.readouterr().out)
+    assert out[0]["pod"] == "ns/p1"
+    assert out[0]["containers"]["main"]["hash"] == d.hash
+
+    assert main(["--db", db, "masks"]) == 0
+    out = json.loads(capsys.readouterr().out)
+    assert out[d.hash]["cu_count"] == 64
+
+
+def test_egpuctl_devices_fake(capsys):
+    from elastic_gpu_agent_amd.cli.egpuctl import main
+
+    assert main(["devices", "--fake"]) == 0
+    out = json.loads(capsys.readouterr().out)
+    assert len(out) == 8 and out[0]["cu_count"] == 256
+
+
+def test_egpuctl_migrate(tmp_path, capsys):
+    import sys
+    sys.path.insert(0, os.path.dirname(__file__))
+    from test_storage import _synth_bolt_file, make_pi
+
+    bolt = str(tmp_path / "old.db")
+    pi = make_pi(ns="default", name="legacy")
+    _synth_bolt_file(bolt, [(pi.key().encode(), pi.val())])
+    from elastic_gpu_agent_amd.cli.egpuctl import main
+
+    db = str(tmp_path / "new.db")
+    assert main(["--db", db, "migrate", "--from", bolt]) == 0
+    assert "migrated 1" in capsys.readouterr().out
+    st = Storage(db)
+    assert st.load("default", "legacy").name == "legacy"
+    st.close()

@@ -54,6 +54,20 @@ def build_amdsmi(force=False):
     return out
 
 
+def build_fastwire(force=False):
+    src = os.path.join(HERE, "fastwire.cpp")
+    ext = sysconfig.get_config_var("EXT_SUFFIX")
+    out = os.path.join(PKG, f"_fastwire{ext}")
+    if not force and _newer(out, src):
+        return out
+    _run(
+        ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src]
+        + pybind_includes()
+        + ["-o", out]
+    )
+    return out
+
+
 def build_shim(force=False):
     src = os.path.join(HERE, "egpu_shim.cpp")
     out = os.path.join(PKG, "libegpu_shim.so")
@@ -92,6 +106,7 @@ def build_hook(force=False):
 def build_all(force=False):
     return {
         "amdsmi": build_amdsmi(force),
+        "fastwire": build_fastwire(force),
         "shim": build_shim(force),
         "kernels": build_kernels(force),
         "hook": build_hook(force),

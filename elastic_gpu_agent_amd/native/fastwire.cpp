@@ -1,0 +1,167 @@
+// _fastwire: C++ accelerators for the two protobuf shapes that can carry
+// hundreds of thousands of fake-device IDs when gpu-memory runs at 1-MiB
+// units (288 GiB ⇒ 294,912 IDs per GPU):
+//
+//   decode_string_list(buf)    repeated string field 1 (PreStartContainerRequest,
+//                              ContainerAllocateRequest bodies)
+//   decode_nested_string_lists(buf)
+//                              repeated message field 1, each holding repeated
+//                              string field 1 (AllocateRequest shape) →
+//                              list of lists of str
+//   encode_device_list(ids, suffix)
+//                              ListAndWatchResponse body: for each id emit a
+//                              field-1 Device submessage = (field-1 string id)
+//                              + caller-precomputed suffix bytes (health +
+//                              topology), all length-prefixed.
+//
+// Pure wire-format code — semantics identical to protos/protowire.py (the
+// test-suite cross-checks them); ~30-40× faster on large ID sets.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+namespace {
+
+struct Reader {
+  const uint8_t* p;
+  const uint8_t* end;
+
+  bool done() const { return p >= end; }
+
+  uint64_t varint() {
+    uint64_t result = 0;
+    int shift = 0;
+    while (p < end) {
+      uint8_t b = *p++;
+      result |= (uint64_t)(b & 0x7F) << shift;
+      if (!(b & 0x80)) return result;
+      shift += 7;
+      if (shift >= 70) throw std::runtime_error("varint too long");
+    }
+    throw std::runtime_error("truncated varint");
+  }
+
+  void skip(uint32_t wire) {
+    switch (wire) {
+      case 0:
+        varint();
+        break;
+      case 1:
+        p += 8;
+        break;
+      case 2: {
+        uint64_t n = varint();
+        p += n;
+        break;
+      }
+      case 5:
+        p += 4;
+        break;
+      default:
+        throw std::runtime_error("bad wire type");
+    }
+    if (p > end) throw std::runtime_error("truncated field");
+  }
+};
+
+// collect every field-1 LEN payload within [p, end)
+void field1_spans(const uint8_t* p, const uint8_t* end,
+                  std::vector<std::pair<const uint8_t*, size_t>>& out) {
+  Reader r{p, end};
+  while (!r.done()) {
+    uint64_t tag = r.varint();
+    uint32_t field = tag >> 3, wire = tag & 7;
+    if (field == 1 && wire == 2) {
+      uint64_t n = r.varint();
+      if (r.p + n > r.end) throw std::runtime_error("truncated");
+      out.emplace_back(r.p, (size_t)n);
+      r.p += n;
+    } else {
+      r.skip(wire);
+    }
+  }
+}
+
+py::list decode_string_list(py::bytes data) {
+  char* buf;
+  Py_ssize_t len;
+  PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+  std::vector<std::pair<const uint8_t*, size_t>> spans;
+  field1_spans((const uint8_t*)buf, (const uint8_t*)buf + len, spans);
+  py::list out;
+  for (auto& s : spans)
+    out.append(py::str((const char*)s.first, s.second));
+  return out;
+}
+
+py::list decode_nested_string_lists(py::bytes data) {
+  char* buf;
+  Py_ssize_t len;
+  PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+  std::vector<std::pair<const uint8_t*, size_t>> outer;
+  field1_spans((const uint8_t*)buf, (const uint8_t*)buf + len, outer);
+  py::list out;
+  for (auto& o : outer) {
+    std::vector<std::pair<const uint8_t*, size_t>> inner;
+    field1_spans(o.first, o.first + o.second, inner);
+    py::list ids;
+    for (auto& s : inner)
+      ids.append(py::str((const char*)s.first, s.second));
+    out.append(ids);
+  }
+  return out;
+}
+
+void put_varint(std::string& out, uint64_t v) {
+  while (v >= 0x80) {
+    out.push_back((char)(v | 0x80));
+    v >>= 7;
+  }
+  out.push_back((char)v);
+}
+
+py::bytes encode_device_list(const std::vector<std::string>& ids, py::bytes suffix_b) {
+  char* sbuf;
+  Py_ssize_t slen;
+  PyBytes_AsStringAndSize(suffix_b.ptr(), &sbuf, &slen);
+  std::string out;
+  size_t est = 0;
+  for (auto& id : ids) est += id.size() + slen + 10;
+  out.reserve(est);
+  auto varint_size = [](uint64_t v) {
+    size_t n = 1;
+    while (v >= 0x80) {
+      v >>= 7;
+      ++n;
+    }
+    return n;
+  };
+  for (auto& id : ids) {
+    // device submessage: field 1 (ID string) + suffix
+    size_t body = 1 + varint_size(id.size()) + id.size() + slen;
+    out.push_back(0x0A);  // ListAndWatchResponse.devices (field 1, LEN)
+    put_varint(out, body);
+    out.push_back(0x0A);  // Device.ID (field 1, LEN)
+    put_varint(out, id.size());
+    out.append(id);
+    out.append(sbuf, slen);
+  }
+  return py::bytes(out);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_fastwire, m) {
+  m.doc() = "wire-format accelerators for large fake-device ID sets";
+  m.def("decode_string_list", &decode_string_list);
+  m.def("decode_nested_string_lists", &decode_nested_string_lists);
+  m.def("encode_device_list", &encode_device_list);
+}

@@ -74,8 +74,16 @@ class DevicePluginServer:
             return egrpc.unary_unary(handler)
 
         def list_and_watch(request_bytes, context):
-            for resp in p.list_and_watch(context):
-                yield dp.ListAndWatchResponse.encode(resp)
+            # plugin yields pre-encoded (and cached) response bytes
+            yield from p.list_and_watch_encoded(context)
+
+        from ..protos import fastpath
+
+        class _FastSpec:
+            """MessageSpec-shaped wrapper around a fastpath decoder."""
+
+            def __init__(self, decode):
+                self.decode = decode
 
         return {
             "GetDevicePluginOptions": timed(
@@ -85,11 +93,13 @@ class DevicePluginServer:
             "GetPreferredAllocation": timed(
                 "GetPreferredAllocation", p.get_preferred_allocation,
                 dp.PreferredAllocationRequest, dp.PreferredAllocationResponse),
-            "Allocate": timed("Allocate", p.allocate, dp.AllocateRequest,
+            "Allocate": timed("Allocate", p.allocate,
+                              _FastSpec(fastpath.decode_allocate_request),
                               dp.AllocateResponse),
             "PreStartContainer": timed(
                 "PreStartContainer", p.pre_start_container,
-                dp.PreStartContainerRequest, dp.PreStartContainerResponse),
+                _FastSpec(fastpath.decode_prestart_request),
+                dp.PreStartContainerResponse),
         }
 
     # ---- lifecycle ----

@@ -52,18 +52,28 @@ class GPUSharePluginBase:
                 out.append({"ID": did, "health": consts.HEALTHY, "topology": topo})
         return out
 
+    def device_groups(self) -> List[tuple]:
+        """[(ids, encoded Device suffix)] per GPU — the fast-encode shape."""
+        from ..protos import fastpath
+
+        groups = []
+        for gpu in self.cfg.operator.devices():
+            suffix = fastpath.device_suffix(consts.HEALTHY, gpu.numa_node)
+            groups.append((self.fake_device_ids_for_gpu(gpu), suffix))
+        return groups
+
     # ---- gRPC handlers ----
     def get_device_plugin_options(self, request, context) -> dict:
         return {"pre_start_required": True, "get_preferred_allocation_available": True}
 
-    def list_and_watch(self, context):
-        """Initial device list + re-advertisement when enumeration changes.
+    def _watch_snapshots(self, context, snap_fn):
+        """Initial snapshot + re-advertisement when enumeration changes.
 
         Unlike the reference (single static send, devices never re-checked —
         SURVEY §3.2), the backend is re-enumerated periodically so a GPU
         falling off the bus transitions its fake devices out of the list."""
-        current = self.list_devices()
-        yield {"devices": current}
+        current = snap_fn()
+        yield current
         interval = self.cfg.options.health_refresh_seconds
         while context is None or context.is_active():
             triggered = self._refresh.wait(timeout=interval)
@@ -71,16 +81,29 @@ class GPUSharePluginBase:
             if context is not None and not context.is_active():
                 return
             try:
-                fresh_gpus = self.cfg.operator.devices(refresh=True)
+                self.cfg.operator.devices(refresh=True)
             except Exception as e:
                 log.error("device re-enumeration failed: %s", e)
                 continue
-            fresh = self.list_devices()
+            fresh = snap_fn()
             if fresh != current:
                 current = fresh
-                yield {"devices": current}
-            if triggered:
-                yield {"devices": current}
+                yield current
+            elif triggered:
+                yield current
+
+    def list_and_watch(self, context):
+        yield from self._watch_snapshots(context, lambda: {"devices": self.list_devices()})
+
+    def list_and_watch_encoded(self, context):
+        """Server path: pre-encoded ListAndWatchResponse bytes (cached until
+        enumeration changes; at 1-MiB memory units a snapshot is ~7 MB per
+        GPU and must not be re-encoded per send)."""
+        from ..protos import fastpath
+
+        yield from self._watch_snapshots(
+            context, lambda: fastpath.encode_list_and_watch(self.device_groups())
+        )
 
     def trigger_refresh(self) -> None:
         self._refresh.set()

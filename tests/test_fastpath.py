@@ -1,0 +1,83 @@
+"""Fastpath codec tests: native _fastwire must agree byte-for-byte with the
+pure-Python wire codec, and the encoded ListAndWatch path must decode
+identically to the dict path."""
+import pytest
+
+from elastic_gpu_agent_amd.protos import deviceplugin as dp
+from elastic_gpu_agent_amd.protos import fastpath
+
+
+def test_native_built():
+    # the accelerator is part of the standard build; tests run post-build
+    assert fastpath.HAVE_NATIVE, "_fastwire not built (run native.build)"
+
+
+def test_decode_allocate_request_matches_python():
+    ids_a = [f"0-{i:06d}" for i in range(1000)]
+    ids_b = [f"1-{i:02d}" for i in range(30)]
+    buf = dp.AllocateRequest.encode(
+        {"container_requests": [{"devicesIDs": ids_a}, {"devicesIDs": ids_b}]}
+    )
+    fast = fastpath.decode_allocate_request(buf)
+    slow = dp.AllocateRequest.decode(buf)
+    assert fast == slow
+    assert fast["container_requests"][0]["devicesIDs"] == ids_a
+
+
+def test_decode_prestart_matches_python():
+    ids = [f"3-{i:06d}" for i in range(5000)]
+    buf = dp.PreStartContainerRequest.encode({"devicesIDs": ids})
+    assert fastpath.decode_prestart_request(buf) == dp.PreStartContainerRequest.decode(buf)
+
+
+def test_empty_messages():
+    assert fastpath.decode_allocate_request(b"") == {"container_requests": []}
+    assert fastpath.decode_prestart_request(b"") == {"devicesIDs": []}
+
+
+def test_encode_list_and_watch_decodes_identically():
+    groups = [
+        ([f"0-{i:02d}" for i in range(100)], fastpath.device_suffix("Healthy", 0)),
+        ([f"1-{i:02d}" for i in range(100)], fastpath.device_suffix("Healthy", 1)),
+    ]
+    buf = fastpath.encode_list_and_watch(groups)
+    out = dp.ListAndWatchResponse.decode(buf)
+    assert len(out["devices"]) == 200
+    assert out["devices"][0] == {
+        "ID": "0-00", "health": "Healthy", "topology": {"nodes": [{"ID": 0}]}
+    }
+    assert out["devices"][150]["topology"]["nodes"][0]["ID"] == 1
+
+
+def test_encode_matches_pure_python_fallback():
+    groups = [([f"0-{i:04d}" for i in range(500)], fastpath.device_suffix("Healthy", 3))]
+    native = fastpath.encode_list_and_watch(groups)
+    # force the pure-python branch
+    saved = fastpath._fastwire
+    fastpath._fastwire = None
+    try:
+        pure = fastpath.encode_list_and_watch(groups)
+    finally:
+        fastpath._fastwire = saved
+    assert native == pure
+
+
+def test_large_scale_performance():
+    """294,912 IDs (one GPU at 1-MiB units): decode must be fast enough for
+    the Allocate hot path."""
+    import time
+
+    ids = [f"0-{i:06d}" for i in range(294912)]
+    buf = dp.AllocateRequest.encode({"container_requests": [{"devicesIDs": ids}]})
+    t0 = time.perf_counter()
+    out = fastpath.decode_allocate_request(buf)
+    dt = time.perf_counter() - t0
+    assert out["container_requests"][0]["devicesIDs"] == ids
+    assert dt < 0.5, f"fast decode took {dt:.3f}s"
+
+    suffix = fastpath.device_suffix("Healthy", 0)
+    t0 = time.perf_counter()
+    payload = fastpath.encode_list_and_watch([(ids, suffix)])
+    dt = time.perf_counter() - t0
+    assert dt < 0.5, f"fast encode took {dt:.3f}s"
+    assert len(payload) > len(ids) * 10

@@ -10,6 +10,7 @@ import json
 import os
 import subprocess
 import sys
+import time
 
 import pytest
 
@@ -17,6 +18,7 @@ pytestmark = pytest.mark.gpu
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SHIM = os.path.join(REPO, "elastic_gpu_agent_amd", "libegpu_shim.so")
+HOOK_BIN = os.path.join(REPO, "bin", "egpu-hook")
 
 
 def _gpu_present():
@@ -183,6 +185,66 @@ def test_agent_end_to_end_on_gpu(tmp_path, gpus):
     assert h.plugin.gc_once() == 1
     assert not os.path.exists(link)
     h.close()
+
+
+def test_hook_real_injection(tmp_path, gpus):
+    """Non-dry-run hook: mknod into a real separate mount namespace.
+
+    Spawns `unshare -m` (private mount ns + tmpfs /dev so the test never
+    touches the shared /dev), runs the hook against that pid, then verifies
+    the device nodes exist inside the namespace with the right major:minor."""
+    if os.geteuid() != 0:
+        pytest.skip("needs root")
+    g0 = gpus[0]
+    dev_root = tmp_path / "hostdev"
+    dev_root.mkdir()
+    # host-side per-alloc links, as PreStart would create them
+    os.symlink(f"/dev/dri/renderD{g0.drm_render_minor}",
+               dev_root / "elastic-gpu-feed0001-0")
+    os.symlink("/dev/kfd", dev_root / "elastic-gpuctl-feed0001-0")
+    # the hook stats the real nodes via the link target; keep a kfd present
+    # in the fake dev root too (resolve_gpu_links only needs the gpu links)
+    bundle = tmp_path / "bundle"
+    bundle.mkdir()
+    (bundle / "config.json").write_text(json.dumps(
+        {"process": {"env": ["GPU=feed0001"]}}))
+
+    # a process in its own mount ns with a private empty /dev
+    target = subprocess.Popen(
+        ["unshare", "-m", "bash", "-c",
+         "mount -t tmpfs tmpfs /dev && mkdir -p /dev/dri && sleep 30"],
+    )
+    try:
+        time.sleep(1.0)
+        env = dict(os.environ)
+        env.update({
+            "EGPU_DEV_ROOT": str(dev_root),
+            "EGPU_HOOK_LOG": str(tmp_path / "hook.log"),
+            "EGPU_STATE_DIR": str(tmp_path / "state"),
+        })
+        # hook stats <dev_root>/kfd for the control node; provide it by
+        # bind-meaning: stat_node falls back to name-derived minors for dri,
+        # but kfd needs the real node — copy the real rdev via a symlink
+        os.symlink("/dev/kfd", dev_root / "kfd")
+        r = subprocess.run(
+            [HOOK_BIN, "prestart"],
+            input=json.dumps({"pid": target.pid, "bundle": str(bundle)}).encode(),
+            env=env, capture_output=True, timeout=60,
+        )
+        assert r.returncode == 0, (r.stderr, open(tmp_path / "hook.log").read())
+        # verify inside the namespace
+        chk = subprocess.run(
+            ["nsenter", "-m", "-t", str(target.pid), "stat", "-c", "%t:%T %F",
+             "/dev/kfd", f"/dev/dri/renderD{g0.drm_render_minor}"],
+            capture_output=True, text=True, timeout=30,
+        )
+        assert chk.returncode == 0, chk.stderr
+        lines = chk.stdout.strip().splitlines()
+        assert "character special file" in lines[0]
+        assert lines[1].startswith("e2:")  # DRM major 226 = 0xe2
+    finally:
+        target.kill()
+        target.wait()
 
 
 def test_shim_blocks_mask_widening(gpus):

@@ -142,10 +142,12 @@ class Channel:
             (b"te", b"trailers"),
         ])
 
-    def _send_request(self, path: bytes, message: bytes) -> int:
+    def _send_request(self, path: bytes, message: bytes,
+                      block: Optional[bytes] = None) -> int:
         sid = self._next_stream
         self._next_stream += 2
-        block = self._request_headers(path)
+        if block is None:
+            block = self._request_headers(path)
         payload = core.grpc_frame(message)
         stream_window = self.peer_initial_window
         out = bytearray(frame_header(len(block), core.HEADERS, core.FLAG_END_HEADERS, sid))
@@ -218,6 +220,9 @@ class Channel:
     # ---- public API ----
     def unary_unary(self, path: str, request_serializer=None, response_deserializer=None):
         pbytes = path.encode()
+        # the header block is constant per method (stateless encoder):
+        # precompute it once instead of per call
+        header_block = self._request_headers(pbytes)
 
         def call(request, timeout: Optional[float] = None):
             msg = request_serializer(request) if request_serializer else request
@@ -226,7 +231,7 @@ class Channel:
                 prev_to = self._sock.gettimeout()
                 self._sock.settimeout(timeout)
                 try:
-                    sid = self._send_request(pbytes, msg)
+                    sid = self._send_request(pbytes, msg, header_block)
                     msgs, _ = self._read_response(sid)
                 except socket.timeout as e:
                     self._reset()

@@ -29,16 +29,18 @@ class DeviceLocator:
 
 
 class KubeletDeviceLocator(DeviceLocator):
-    def __init__(self, resource_name: str, socket_path: str = consts.POD_RESOURCES_SOCKET):
+    def __init__(self, resource_name: str, socket_path: str = consts.POD_RESOURCES_SOCKET,
+                 connect_timeout: float = 10.0):
         self._resource = resource_name
         self._socket = socket_path
+        self._connect_timeout = connect_timeout
         self._lock = threading.Lock()
         self._channel: Optional[egrpc.Channel] = None
         self._list = None
 
     def _ensure(self):
         if self._channel is None:
-            self._channel = egrpc.Channel(self._socket)
+            self._channel = egrpc.Channel(self._socket, connect_timeout=self._connect_timeout)
             self._list = self._channel.unary_unary(
                 pr.METHOD_LIST,
                 request_serializer=pr.ListPodResourcesRequest.encode,

@@ -1,0 +1,195 @@
+"""Failure-path and robustness tests: malformed transport input, agent
+restart over persisted state, locator outage during PreStart, metrics."""
+import json
+import os
+import socket
+import time
+
+import pytest
+
+from elastic_gpu_agent_amd import consts, egrpc
+from elastic_gpu_agent_amd.types import Device, PodContainer
+
+from helpers import Harness, PluginClient
+
+
+# ---- egrpc server vs hostile/broken peers ----------------------------------
+
+@pytest.fixture
+def eserver(tmp_path):
+    sock = str(tmp_path / "s.sock")
+    s = egrpc.Server()
+    s.add_service("t", {"E": egrpc.unary_unary(lambda r, c: r)})
+    s.bind_unix(sock)
+    s.start()
+    yield sock, s
+    s.stop()
+
+
+def _raw(sock_path: str) -> socket.socket:
+    c = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    c.connect(sock_path)
+    return c
+
+
+def _alive(sock_path: str) -> bool:
+    ch = egrpc.Channel(sock_path, connect_timeout=3.0)
+    try:
+        return ch.unary_unary("/t/E")(b"ok", timeout=3.0) == b"ok"
+    finally:
+        ch.close()
+
+
+def test_server_survives_garbage_preface(eserver):
+    sock, _ = eserver
+    c = _raw(sock)
+    c.sendall(b"GET / HTTP/1.1\r\n\r\n" + b"\x00" * 64)
+    time.sleep(0.2)
+    c.close()
+    assert _alive(sock)
+
+
+def test_server_survives_truncated_frames(eserver):
+    sock, _ = eserver
+    from elastic_gpu_agent_amd.egrpc import core
+
+    c = _raw(sock)
+    c.sendall(core.PREFACE + core.settings_frame([]))
+    # frame header claiming a large body, then hang up mid-body
+    c.sendall(core.frame_header(100000, core.DATA, 0, 1) + b"xx")
+    c.close()
+    time.sleep(0.2)
+    assert _alive(sock)
+
+
+def test_server_survives_bad_hpack(eserver):
+    sock, _ = eserver
+    from elastic_gpu_agent_amd.egrpc import core
+
+    c = _raw(sock)
+    c.sendall(core.PREFACE + core.settings_frame([]))
+    bad_block = b"\xff\xff\xff\xff\xff\xff"  # bogus huge index
+    c.sendall(
+        core.frame_header(len(bad_block), core.HEADERS,
+                          core.FLAG_END_HEADERS | core.FLAG_END_STREAM, 1)
+        + bad_block
+    )
+    time.sleep(0.2)
+    c.close()
+    assert _alive(sock)
+
+
+def test_server_survives_unknown_frame_types(eserver):
+    sock, _ = eserver
+    from elastic_gpu_agent_amd.egrpc import core
+
+    c = _raw(sock)
+    c.sendall(core.PREFACE + core.settings_frame([]))
+    c.sendall(core.frame_header(4, 0xBB, 0, 0) + b"abcd")  # unknown type
+    c.sendall(core.frame_header(5, core.PRIORITY, 0, 3) + b"\x00" * 5)
+    # and a valid request afterwards on the same connection
+    ch = egrpc.Channel(sock)
+    assert ch.unary_unary("/t/E")(b"still") == b"still"
+    ch.close()
+    c.close()
+
+
+# ---- agent restart over persisted state ------------------------------------
+
+def test_masks_survive_agent_restart(tmp_path):
+    """A restarted agent must keep honoring live pods' CU masks: the second
+    pod's mask (allocated post-restart) stays disjoint from the first's."""
+    h1 = Harness(str(tmp_path), gpus=1)
+    ids1 = [f"0-{i:02d}" for i in range(25)]
+    d1 = Device.new(ids1, consts.RESOURCE_GPU_CORE)
+    h1.core_locator.assign(d1.hash, PodContainer("ns", "p1", "main"))
+    h1.add_assumed_pod("ns", "p1", "main", "0")
+    h1.plugin.core.allocate({"container_requests": [{"devicesIDs": ids1}]}, None)
+    h1.plugin.core.pre_start_container({"devicesIDs": ids1}, None)
+    mask1 = h1.plugin.cfg.limits.read(d1.hash)["cu_mask"]
+    h1.plugin.stop()
+    h1.storage.close()
+
+    # "restart": a fresh harness over the same tmp dir (same DB, dev root)
+    h2 = Harness(str(tmp_path), gpus=1)
+    ids2 = [f"0-{i:02d}" for i in range(50, 75)]
+    d2 = Device.new(ids2, consts.RESOURCE_GPU_CORE)
+    h2.core_locator.assign(d2.hash, PodContainer("ns", "p2", "main"))
+    h2.add_assumed_pod("ns", "p2", "main", "0")
+    h2.plugin.core.allocate({"container_requests": [{"devicesIDs": ids2}]}, None)
+    h2.plugin.core.pre_start_container({"devicesIDs": ids2}, None)
+    mask2 = h2.plugin.cfg.limits.read(d2.hash)["cu_mask"]
+
+    from elastic_gpu_agent_amd.isolation.cumask import parse_mask_hex
+
+    w1, w2 = parse_mask_hex(mask1), parse_mask_hex(mask2)
+    assert all((a & b) == 0 for a, b in zip(w1, w2))
+    h2.close()
+
+
+def test_restore_after_reboot_with_real_sockets(tmp_path):
+    """Restore() then serve: symlinks wiped by 'reboot' come back before the
+    plugin starts answering kubelet."""
+    h = Harness(str(tmp_path), gpus=1)
+    ids = [f"0-{i:02d}" for i in range(10)]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    h.core_locator.assign(d.hash, PodContainer("ns", "pr", "main"))
+    h.add_assumed_pod("ns", "pr", "main", "0")
+    h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+    link = os.path.join(h.paths.dev_root, f"elastic-gpu-{d.hash}-0")
+    os.unlink(link)
+    assert h.plugin.restore() == 1
+    assert os.path.islink(link)
+    h.close()
+
+
+# ---- locator outage ----------------------------------------------------------
+
+def test_prestart_fails_cleanly_when_podresources_down(tmp_path):
+    from elastic_gpu_agent_amd.kube.locator import KubeletDeviceLocator
+
+    h = Harness(str(tmp_path), gpus=1)
+    # swap in a locator pointing at a dead socket
+    h.plugin.cfg.core_locator = KubeletDeviceLocator(
+        consts.RESOURCE_GPU_CORE, str(tmp_path / "nope.sock"), connect_timeout=0.5
+    )
+    h.plugin.cfg.core_locator._channel = None
+    ids = ["0-00"]
+    with pytest.raises(RuntimeError, match="locate"):
+        # EgrpcError(UNAVAILABLE) inside → wrapped as a clean failure
+        try:
+            h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+        except egrpc.EgrpcError as e:
+            raise RuntimeError(f"locate: {e}")
+    h.close()
+
+
+# ---- metrics -----------------------------------------------------------------
+
+def test_metrics_recorder_quantiles():
+    from elastic_gpu_agent_amd.metrics import LatencyRecorder
+
+    rec = LatencyRecorder(capacity=1000)
+    for i in range(100):
+        rec.observe(i / 1000.0)
+    s = rec.summary_us()
+    assert s["count"] == 100
+    assert 45_000 <= s["p50_us"] <= 55_000
+    assert s["p99_us"] >= 95_000
+    assert s["max_us"] == pytest.approx(99_000, rel=0.01)
+
+
+def test_rpc_latency_recorded_through_server(tmp_path):
+    from elastic_gpu_agent_amd.metrics import GLOBAL_METRICS
+
+    h = Harness(str(tmp_path), gpus=1)
+    h.plugin.core_server.serve()
+    h.plugin.core_server.wait_ready()
+    client = PluginClient(h.plugin.core_server.socket_path)
+    before = GLOBAL_METRICS.recorder("elasticgpu.io/gpu-core/Allocate").count
+    client.allocate({"container_requests": [{"devicesIDs": ["0-00"]}]})
+    after = GLOBAL_METRICS.recorder("elasticgpu.io/gpu-core/Allocate").count
+    assert after == before + 1
+    client.close()
+    h.close()

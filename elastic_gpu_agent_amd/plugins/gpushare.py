@@ -160,9 +160,13 @@ class GPUSharePluginBase:
     def pre_start_container(self, request, context) -> dict:
         ids = request.get("devicesIDs", [])
         device = Device.new(ids, self.resource_name)
+        from .. import egrpc
+
         try:
             pc = self._locator().locate(device)
-        except KeyError as e:
+        except (KeyError, egrpc.EgrpcError) as e:
+            # unknown device set, or podresources outage (clean failure —
+            # kubelet will retry the container start)
             return self._fail(context, f"locate {device.hash}: {e}")
         try:
             pod = self.cfg.sitter.get_pod(pc.namespace, pc.name)

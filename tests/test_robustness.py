@@ -154,14 +154,11 @@ def test_prestart_fails_cleanly_when_podresources_down(tmp_path):
     h.plugin.cfg.core_locator = KubeletDeviceLocator(
         consts.RESOURCE_GPU_CORE, str(tmp_path / "nope.sock"), connect_timeout=0.5
     )
-    h.plugin.cfg.core_locator._channel = None
     ids = ["0-00"]
+    # podresources outage surfaces as a clean handler failure (kubelet
+    # retries), never an unhandled transport exception
     with pytest.raises(RuntimeError, match="locate"):
-        # EgrpcError(UNAVAILABLE) inside → wrapped as a clean failure
-        try:
-            h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
-        except egrpc.EgrpcError as e:
-            raise RuntimeError(f"locate: {e}")
+        h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
     h.close()
 
 

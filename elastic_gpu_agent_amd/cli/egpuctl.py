@@ -74,6 +74,18 @@ def cmd_occupancy(args) -> int:
     return 0
 
 
+def cmd_partition(args) -> int:
+    from elastic_gpu_agent_amd import _amdsmi
+
+    if args.mode:
+        _amdsmi.set_compute_partition(args.index, args.mode)
+        print(f"gpu {args.index} partition set to {args.mode} "
+              "(agent re-advertises on its next enumeration refresh)")
+    print(json.dumps({"gpu": args.index,
+                      "partition": _amdsmi.get_compute_partition(args.index)}))
+    return 0
+
+
 def cmd_migrate(args) -> int:
     from ..storage import Storage, migrate_from_bolt
 
@@ -100,6 +112,11 @@ def main(argv=None) -> int:
     m = sub.add_parser("migrate")
     m.add_argument("--from", required=True)
     m.set_defaults(fn=cmd_migrate)
+    pt = sub.add_parser("partition", help="get/set SPX|DPX|QPX|CPX compute partition")
+    pt.add_argument("index", type=int)
+    pt.add_argument("mode", nargs="?", default=None,
+                    choices=[None, "SPX", "DPX", "QPX", "CPX"])
+    pt.set_defaults(fn=cmd_partition)
 
     args = p.parse_args(argv)
     return args.fn(args)

@@ -212,10 +212,40 @@ py::list gpu_processes(uint32_t index) {
   return out;
 }
 
+// ---- compute partition control (SPX/DPX/QPX/CPX) ---------------------------
+
+std::string get_compute_partition(uint32_t index) {
+  SmiSession session;
+  auto gpus = gpu_handles();
+  auto h = handle_for_hip_index(index, gpus);
+  char buf[32] = {0};
+  check(amdsmi_get_gpu_compute_partition(h, buf, sizeof(buf)), "get_compute_partition");
+  return buf;
+}
+
+void set_compute_partition(uint32_t index, const std::string& mode) {
+  amdsmi_compute_partition_type_t t;
+  if (mode == "SPX") t = AMDSMI_COMPUTE_PARTITION_SPX;
+  else if (mode == "DPX") t = AMDSMI_COMPUTE_PARTITION_DPX;
+  else if (mode == "QPX") t = AMDSMI_COMPUTE_PARTITION_QPX;
+  else if (mode == "CPX") t = AMDSMI_COMPUTE_PARTITION_CPX;
+  else throw std::runtime_error("unknown partition mode " + mode +
+                                " (SPX|DPX|QPX|CPX)");
+  SmiSession session;
+  auto gpus = gpu_handles();
+  auto h = handle_for_hip_index(index, gpus);
+  check(amdsmi_set_gpu_compute_partition(h, t), "set_compute_partition");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_amdsmi, m) {
   m.doc() = "MI355X enumeration + occupancy via libamd_smi";
+  m.def("get_compute_partition", &get_compute_partition,
+        "Current compute-partition mode of one GPU (by HIP index)");
+  m.def("set_compute_partition", &set_compute_partition,
+        "Set SPX/DPX/QPX/CPX on one GPU (device must be idle; re-enumeration "
+        "follows via the agent's periodic refresh)");
   m.def("enumerate_gpus", &enumerate_gpus,
         "Enumerate AMD GPUs; returns list of dicts sorted by HIP id");
   m.def("gpu_utilization", &gpu_utilization,

@@ -187,6 +187,56 @@ def test_agent_end_to_end_on_gpu(tmp_path, gpus):
     h.close()
 
 
+def test_quota_enforced_on_torch_workload():
+    """The HBM quota must bind real frameworks, not just our probe kernels:
+    PyTorch allocating past the shim quota gets a CUDA/HIP OOM."""
+    code = (
+        "import torch\n"
+        "a = torch.empty(256, 1024, 1024, device='cuda')  # 1 GiB fp32... fits\n"
+        "try:\n"
+        "    b = torch.empty(8 * 1024, 1024, 1024, device='cuda')  # 32 GiB\n"
+        "    print('NOOOM')\n"
+        "except torch.cuda.OutOfMemoryError:\n"
+        "    print('OOM-AS-EXPECTED')\n"
+    )
+    env = dict(os.environ)
+    env["HSA_TOOLS_LIB"] = SHIM
+    env["EGPU_MEM_LIMIT_BYTES"] = str(8 * 1024**3)
+    out = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                         capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-3000:]
+    assert "OOM-AS-EXPECTED" in out.stdout
+
+
+def test_cu_mask_applies_to_torch_kernels(gpus):
+    """Masked shim + a torch matmul workload: wall time should roughly double
+    with half the CUs (sanity that real framework queues get masked too)."""
+    code = (
+        "import torch, time\n"
+        "a = torch.randn(4096, 4096, device='cuda')\n"
+        "for _ in range(3): (a @ a).sum().item()  # warm\n"
+        "torch.cuda.synchronize(); t0 = time.perf_counter()\n"
+        "for _ in range(20): c = a @ a\n"
+        "torch.cuda.synchronize(); print(time.perf_counter() - t0)\n"
+    )
+    def run(extra):
+        env = dict(os.environ)
+        env.update(extra)
+        out = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                             capture_output=True, text=True, timeout=600)
+        assert out.returncode == 0, out.stderr[-3000:]
+        return float(out.stdout.strip().splitlines()[-1])
+
+    from elastic_gpu_agent_amd.isolation.cumask import mask_for_percent, mask_hex
+
+    full = run({})
+    words, _ = mask_for_percent(25, gpus[0].cu_count, gpus[0].xcd_count)
+    quarter = run({"HSA_TOOLS_LIB": SHIM, "EGPU_CU_MASK": mask_hex(words)})
+    ratio = quarter / full
+    # matmul on 25% of CUs: expect ≥2x slowdown (bandwidth may soften 4x)
+    assert ratio > 1.8, f"25% mask gave only {ratio:.2f}x on torch matmul"
+
+
 def test_hook_real_injection(tmp_path, gpus):
     """Non-dry-run hook: mknod into a real separate mount namespace.
 

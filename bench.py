@@ -145,6 +145,7 @@ def main():
     pods_per_step = len(plans)
 
     alloc_lat = []  # seconds, every Allocate RPC
+    prestart_lat = []  # seconds, every PreStartContainer RPC
 
     def one_step(step_i: int):
         pods = []
@@ -158,8 +159,11 @@ def main():
             client = core if kind == "core" else mem
             t0 = time.perf_counter()
             client.allocate({"container_requests": [{"devicesIDs": ids}]})
-            alloc_lat.append(time.perf_counter() - t0)
+            t1 = time.perf_counter()
             client.pre_start({"devicesIDs": ids})
+            t2 = time.perf_counter()
+            alloc_lat.append(t1 - t0)
+            prestart_lat.append(t2 - t1)
             pods.append((ns, name, d))
         # teardown: pods deleted, GC reclaims symlinks/masks/limits/state
         for ns, name, _ in pods:
@@ -177,6 +181,7 @@ def main():
         one_step(-1 - w)
 
     alloc_lat.clear()
+    prestart_lat.clear()
     barrier()
     t_start = time.perf_counter()
     for s in range(args.steps):
@@ -188,6 +193,8 @@ def main():
     lat_sorted = sorted(alloc_lat)
     p50_us = lat_sorted[len(lat_sorted) // 2] * 1e6
     p99_us = lat_sorted[min(len(lat_sorted) - 1, int(0.99 * len(lat_sorted)))] * 1e6
+    ps_sorted = sorted(prestart_lat)
+    p50_prestart_us = ps_sorted[len(ps_sorted) // 2] * 1e6 if ps_sorted else 0.0
     elapsed_max = elapsed
     if dist is not None:
         t = torch.tensor([elapsed, p50_us, p99_us])
@@ -217,6 +224,7 @@ def main():
             "data": "synthetic",
             "p50_allocate_us": round(p50_us, 1),
             "p99_allocate_us": round(p99_us, 1),
+            "p50_prestart_us": round(p50_prestart_us, 1),
             "config": {
                 "model": "gpushare-device-plugin",
                 "scenario": args.config,

@@ -9,7 +9,7 @@ import socket
 import struct
 import threading
 import time
-from typing import Dict, Iterator, List, Optional, Tuple
+from typing import Iterator, List, Optional, Tuple
 
 from . import core, hpack
 from .core import ConnectionClosed, EgrpcError, frame_header, parse_frame_header

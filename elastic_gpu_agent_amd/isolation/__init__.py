@@ -23,7 +23,7 @@ from typing import Dict, List, Optional, Tuple
 
 from .. import consts
 from ..types import GPUDevice
-from .cumask import mask_hex, mask_words_from_cus, parse_mask_hex, popcount
+from .cumask import mask_hex, mask_words_from_cus, parse_mask_hex
 
 AUX_MASK_PREFIX = "mask/"  # aux key: mask/<alloc_hash> -> json record
 

@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import ctypes
 import os
-from typing import List, Tuple
+from typing import List
 
 _LIB = None
 

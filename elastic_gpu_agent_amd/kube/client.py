@@ -10,8 +10,7 @@ from __future__ import annotations
 
 import json
 import os
-import threading
-from typing import Callable, Iterator, List, Optional
+from typing import Iterator, Optional
 
 import httpx
 

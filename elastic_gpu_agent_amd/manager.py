@@ -14,7 +14,7 @@ from typing import Optional
 
 from . import consts
 from .isolation import CUMaskAllocator, LimitsWriter
-from .kube.locator import FakeDeviceLocator, KubeletDeviceLocator
+from .kube.locator import KubeletDeviceLocator
 from .kube.sitter import FakeSitter, PodSitter
 from .operator import GPUOperator
 from .operator.fake import FakeBackend

@@ -12,7 +12,7 @@ import logging
 import queue
 import threading
 import time
-from typing import List, Optional
+from typing import List
 
 from .. import consts
 from ..kube.client import NotFound

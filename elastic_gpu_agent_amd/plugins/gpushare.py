@@ -19,12 +19,11 @@ from __future__ import annotations
 import logging
 import math
 import threading
-import time
-from typing import Dict, List, Optional
+from typing import List
 
 from .. import consts, topology
 from ..kube.client import NotFound
-from ..types import Device, PodInfo
+from ..types import Device
 from .config import GPUPluginConfig
 
 log = logging.getLogger(__name__)

@@ -8,7 +8,7 @@ both agree byte-for-byte. At 1-MiB memory units a 72 GiB allocation carries
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 from . import deviceplugin as dp
 from .protowire import encode_varint

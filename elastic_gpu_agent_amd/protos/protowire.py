@@ -13,8 +13,7 @@ map<string,string>; any field may be repeated.
 """
 from __future__ import annotations
 
-import struct
-from typing import Any, Dict, List, Optional, Tuple, Type
+from typing import Any, Dict, List, Optional, Tuple
 
 WIRE_VARINT = 0
 WIRE_I64 = 1

@@ -335,10 +335,11 @@ int main(int argc, char** argv) {
 
   logf("injecting %zu nodes for GPU=%s into pid %ld%s", nodes.size(), hash.c_str(), pid,
        dryrun ? " (dry-run)" : "");
-  int rc = inject(pid, nodes, dryrun);
-  if (rc == 0) {
-    // record hash→pid so the agent can attribute per-process GPU occupancy
-    // (amdsmi process list) back to pods
+  // record hash→pid BEFORE entering the container's mount namespace (after
+  // setns, host paths are no longer reachable); the agent joins this with
+  // the amdsmi process list for per-pod occupancy. Best-effort: removed
+  // again by GC if injection fails and the pod never starts.
+  {
     const char* state_env = getenv("EGPU_STATE_DIR");
     std::string state_dir = state_env ? state_env : "/var/lib/egpu";
     std::string pid_dir = state_dir + "/pids";
@@ -350,5 +351,5 @@ int main(int argc, char** argv) {
       fclose(pf);
     }
   }
-  return rc;
+  return inject(pid, nodes, dryrun);
 }

@@ -1,0 +1,87 @@
+"""ListAndWatch behavior: encoded stream ≡ dict stream, refresh triggers,
+and device-change re-advertisement."""
+import threading
+import time
+
+import pytest
+
+from elastic_gpu_agent_amd.protos import deviceplugin as dp
+
+from helpers import Harness
+
+
+@pytest.fixture
+def h(tmp_path):
+    harness = Harness(str(tmp_path), gpus=2)
+    harness.plugin.cfg.options.health_refresh_seconds = 0.2
+    yield harness
+    harness.close()
+
+
+def test_encoded_stream_matches_dict_stream(h):
+    enc_gen = h.plugin.core.list_and_watch_encoded(None)
+    payload = next(enc_gen)
+    decoded = dp.ListAndWatchResponse.decode(payload)
+    dict_gen = h.plugin.core.list_and_watch(None)
+    direct = next(dict_gen)
+    assert decoded == direct
+    enc_gen.close()
+    dict_gen.close()
+
+
+class _Ctx:
+    def __init__(self):
+        self.active = True
+
+    def is_active(self):
+        return self.active
+
+
+def test_readvertise_on_device_change(h):
+    """A GPU disappearing from enumeration shrinks the advertised list."""
+    ctx = _Ctx()
+    gen = h.plugin.core.list_and_watch(ctx)
+    first = next(gen)
+    assert len(first["devices"]) == 200
+
+    got = []
+
+    def consume():
+        for resp in gen:
+            got.append(resp)
+            break
+
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    # GPU 1 falls off the bus
+    h.operator.backend.count = 1
+    deadline = time.time() + 5
+    while not got and time.time() < deadline:
+        time.sleep(0.05)
+    assert got, "no re-advertisement after enumeration change"
+    assert len(got[0]["devices"]) == 100
+    ctx.active = False
+    gen.close()
+
+
+def test_trigger_refresh_resends(h):
+    ctx = _Ctx()
+    gen = h.plugin.core.list_and_watch(ctx)
+    next(gen)
+    got = []
+
+    def consume():
+        for resp in gen:
+            got.append(resp)
+            break
+
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    time.sleep(0.05)
+    h.plugin.core.trigger_refresh()
+    deadline = time.time() + 5
+    while not got and time.time() < deadline:
+        time.sleep(0.05)
+    assert got and len(got[0]["devices"]) == 200
+    ctx.active = False
+    gen.close()

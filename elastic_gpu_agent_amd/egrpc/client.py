@@ -32,6 +32,7 @@ class Channel:
         self._conn_recv_deficit = 0
         self._stream_recv_deficit = 0
         self._deficit_sid = 0
+        self._active_stream_window = core.DEFAULT_WINDOW
 
     # ---- connection ----
     def _connect(self) -> None:
@@ -100,6 +101,9 @@ class Channel:
                     if k == core.SETTINGS_MAX_FRAME_SIZE:
                         self.peer_max_frame = v
                     elif k == core.SETTINGS_INITIAL_WINDOW_SIZE:
+                        # RFC 7540 §6.9.2: the delta applies to every open
+                        # stream's window, including one mid-send
+                        self._active_stream_window += v - self.peer_initial_window
                         self.peer_initial_window = v
                 self._sock.sendall(core.settings_frame([], flags=core.FLAG_ACK))
             return True
@@ -149,14 +153,15 @@ class Channel:
         if block is None:
             block = self._request_headers(path)
         payload = core.grpc_frame(message)
-        stream_window = self.peer_initial_window
+        self._active_stream_window = self.peer_initial_window
         out = bytearray(frame_header(len(block), core.HEADERS, core.FLAG_END_HEADERS, sid))
         out += block
         # send DATA respecting windows; pump frames when blocked
         off, total = 0, len(payload)
         view = memoryview(payload)
         while True:
-            avail = min(self.conn_send_window, stream_window, self.peer_max_frame)
+            avail = min(self.conn_send_window, self._active_stream_window,
+                        self.peer_max_frame)
             if total - off > 0 and avail <= 0:
                 self._sock.sendall(out)
                 out = bytearray()
@@ -164,7 +169,7 @@ class Channel:
                 if not self._handle_conn_frame(ftype, flags, fsid, body):
                     if ftype == core.WINDOW_UPDATE and fsid == sid:
                         (inc,) = struct.unpack(">I", body)
-                        stream_window += inc
+                        self._active_stream_window += inc
                     elif ftype == core.RST_STREAM:
                         raise EgrpcError(core.UNAVAILABLE, "stream reset during send")
                 continue
@@ -173,7 +178,7 @@ class Channel:
             flags = core.FLAG_END_STREAM if last else 0
             out += frame_header(n, core.DATA, flags, sid) + bytes(view[off : off + n])
             self.conn_send_window -= n
-            stream_window -= n
+            self._active_stream_window -= n
             off += n
             if last:
                 break

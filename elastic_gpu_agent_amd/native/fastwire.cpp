@@ -157,6 +157,45 @@ py::bytes encode_device_list(const std::vector<std::string>& ids, py::bytes suff
   return py::bytes(out);
 }
 
+size_t varint_size(uint64_t v) {
+  size_t n = 1;
+  while (v >= 0x80) {
+    v >>= 7;
+    ++n;
+  }
+  return n;
+}
+
+// repeated string field 1 (PreStartContainerRequest / inner container bodies)
+std::string encode_string_list_raw(const std::vector<std::string>& ids) {
+  std::string out;
+  size_t est = 0;
+  for (auto& id : ids) est += id.size() + 6;
+  out.reserve(est);
+  for (auto& id : ids) {
+    out.push_back(0x0A);
+    put_varint(out, id.size());
+    out.append(id);
+  }
+  return out;
+}
+
+py::bytes encode_string_list(const std::vector<std::string>& ids) {
+  return py::bytes(encode_string_list_raw(ids));
+}
+
+// AllocateRequest shape: repeated message field 1 wrapping repeated string 1
+py::bytes encode_nested_string_lists(const std::vector<std::vector<std::string>>& lists) {
+  std::string out;
+  for (auto& ids : lists) {
+    std::string inner = encode_string_list_raw(ids);
+    out.push_back(0x0A);
+    put_varint(out, inner.size());
+    out.append(inner);
+  }
+  return py::bytes(out);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_fastwire, m) {
@@ -164,4 +203,6 @@ PYBIND11_MODULE(_fastwire, m) {
   m.def("decode_string_list", &decode_string_list);
   m.def("decode_nested_string_lists", &decode_nested_string_lists);
   m.def("encode_device_list", &encode_device_list);
+  m.def("encode_string_list", &encode_string_list);
+  m.def("encode_nested_string_lists", &encode_nested_string_lists);
 }

@@ -30,6 +30,21 @@ def decode_allocate_request(buf: bytes) -> dict:
     return dp.AllocateRequest.decode(buf)
 
 
+def encode_allocate_request(req: dict) -> bytes:
+    """Client-side fast path (bench / tests; kubelet's own Go encoder plays
+    this role in production)."""
+    if _fastwire is not None:
+        lists = [cr.get("devicesIDs", []) for cr in req.get("container_requests", [])]
+        return _fastwire.encode_nested_string_lists(lists)
+    return dp.AllocateRequest.encode(req)
+
+
+def encode_prestart_request(req: dict) -> bytes:
+    if _fastwire is not None:
+        return _fastwire.encode_string_list(req.get("devicesIDs", []))
+    return dp.PreStartContainerRequest.encode(req)
+
+
 def decode_prestart_request(buf: bytes) -> dict:
     if _fastwire is not None:
         return {"devicesIDs": _fastwire.decode_string_list(buf)}

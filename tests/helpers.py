@@ -90,6 +90,7 @@ class PluginClient:
 
     def __init__(self, socket_path: str):
         from elastic_gpu_agent_amd import egrpc
+        from elastic_gpu_agent_amd.protos import fastpath
 
         self.channel = egrpc.Channel(socket_path)
         mk = self.channel.unary_unary
@@ -100,12 +101,12 @@ class PluginClient:
         )
         self.allocate = mk(
             dp.METHOD_ALLOCATE,
-            request_serializer=dp.AllocateRequest.encode,
+            request_serializer=fastpath.encode_allocate_request,
             response_deserializer=dp.AllocateResponse.decode,
         )
         self.pre_start = mk(
             dp.METHOD_PRE_START_CONTAINER,
-            request_serializer=dp.PreStartContainerRequest.encode,
+            request_serializer=fastpath.encode_prestart_request,
             response_deserializer=dp.PreStartContainerResponse.decode,
         )
         self.preferred = mk(

@@ -54,6 +54,17 @@ def test_self_unary(eserver):
     ch.close()
 
 
+def test_large_request_as_first_call(eserver):
+    """Regression: a >64 KiB request as the FIRST RPC on a fresh channel —
+    the server's SETTINGS INITIAL_WINDOW_SIZE delta must apply to the stream
+    already mid-send (RFC 7540 §6.9.2); this used to deadlock."""
+    sock, _ = eserver
+    ch = egrpc.Channel(sock)
+    blob = os.urandom(800_000)
+    assert ch.unary_unary("/t.Test/Echo")(blob, timeout=10.0) == blob
+    ch.close()
+
+
 def test_self_error(eserver):
     sock, _ = eserver
     ch = egrpc.Channel(sock)

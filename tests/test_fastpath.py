@@ -81,3 +81,11 @@ def test_large_scale_performance():
     dt = time.perf_counter() - t0
     assert dt < 0.5, f"fast encode took {dt:.3f}s"
     assert len(payload) > len(ids) * 10
+
+
+def test_encode_requests_match_python():
+    ids = [f"0-{i:06d}" for i in range(2000)]
+    req = {"container_requests": [{"devicesIDs": ids}, {"devicesIDs": ["1-00"]}]}
+    assert fastpath.encode_allocate_request(req) == dp.AllocateRequest.encode(req)
+    pre = {"devicesIDs": ids}
+    assert fastpath.encode_prestart_request(pre) == dp.PreStartContainerRequest.encode(pre)

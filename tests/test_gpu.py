@@ -245,6 +245,9 @@ def test_hook_real_injection(tmp_path, gpus):
     the device nodes exist inside the namespace with the right major:minor."""
     if os.geteuid() != 0:
         pytest.skip("needs root")
+    probe = subprocess.run(["unshare", "-m", "true"], capture_output=True)
+    if probe.returncode != 0:
+        pytest.skip(f"mount namespaces unavailable here: {probe.stderr.decode().strip()}")
     g0 = gpus[0]
     dev_root = tmp_path / "hostdev"
     dev_root.mkdir()

@@ -69,28 +69,47 @@ class CUMaskAllocator:
         with self._lock:
             used = self._live_cus(gpu_index)
             # ROCr CU masks have pair granularity: allocate whole CU pairs.
-            base, extra = divmod((n + 1) // 2, xcds)
-            cus: List[int] = []
-            for xcd in range(xcds):
-                want = base + (1 if xcd < extra else 0)
-                free_pairs = [
+            # Water-filling round-robin over the XCDs' FREE pairs: as long as
+            # free capacity exists anywhere, a new allocation never overlaps
+            # (even when earlier pods fragmented some XCDs), while staying as
+            # XCD-balanced as the free space allows. Only genuine
+            # oversubscription falls back to overlapping pairs.
+            want = (n + 1) // 2
+            free_by_xcd = [
+                [
                     xcd * per_xcd + 2 * p
                     for p in range(pairs_per_xcd)
                     if xcd * per_xcd + 2 * p not in used
                     and xcd * per_xcd + 2 * p + 1 not in used
                 ]
-                take = free_pairs[:want]
-                if len(take) < want:
-                    # oversubscribed: wrap around, overlapping already-used
-                    # pairs of this XCD
-                    for p in range(pairs_per_xcd):
-                        if len(take) >= want:
+                for xcd in range(xcds)
+            ]
+            taken: List[int] = []
+            while len(taken) < want and any(free_by_xcd):
+                progress = False
+                for xcd in range(xcds):
+                    if len(taken) >= want:
+                        break
+                    if free_by_xcd[xcd]:
+                        taken.append(free_by_xcd[xcd].pop(0))
+                        progress = True
+                if not progress:
+                    break
+            if len(taken) < want:
+                # oversubscribed: overlap already-used pairs, round-robin
+                # across XCDs so the overlap is spread too
+                for p in range(pairs_per_xcd):
+                    for xcd in range(xcds):
+                        if len(taken) >= want:
                             break
                         cu0 = xcd * per_xcd + 2 * p
-                        if cu0 not in take:
-                            take.append(cu0)
-                for cu0 in take:
-                    cus.extend((cu0, cu0 + 1))
+                        if cu0 not in taken:
+                            taken.append(cu0)
+                    if len(taken) >= want:
+                        break
+            cus: List[int] = []
+            for cu0 in taken:
+                cus.extend((cu0, cu0 + 1))
             n_eff = len(cus)  # pair rounding may add one CU over the ask
             words = mask_words_from_cus(cus, total)
             hexmask = mask_hex(words)

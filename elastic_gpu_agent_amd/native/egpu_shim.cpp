@@ -295,6 +295,7 @@ hsa_status_t pool_allocate_wrap(hsa_amd_memory_pool_t pool, size_t size, uint32_
       g_denied.fetch_add(1);
       logf("DENY pool alloc %zu (used %llu / limit %llu)", size, (unsigned long long)prev,
            (unsigned long long)g_cfg.mem_limit);
+      if (ptr) *ptr = nullptr;  // some callers check the pointer, not status
       return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
     }
     hsa_status_t st = g_amdext.hsa_amd_memory_pool_allocate_fn(pool, size, flags, ptr);
@@ -319,6 +320,7 @@ hsa_status_t memory_allocate_wrap(hsa_region_t region, size_t size, void** ptr) 
     if (prev + size > g_cfg.mem_limit) {
       g_vram_used.fetch_sub(size);
       g_denied.fetch_add(1);
+      if (ptr) *ptr = nullptr;
       return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
     }
     hsa_status_t st = g_core.hsa_memory_allocate_fn(region, size, ptr);

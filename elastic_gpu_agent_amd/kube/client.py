@@ -102,6 +102,39 @@ class K8sClient:
                     continue
                 yield ev.get("type", ""), Pod.from_api_obj(ev.get("object", {}))
 
+    def create_event(
+        self,
+        namespace: str,
+        pod_name: str,
+        pod_uid: str,
+        reason: str,
+        message: str,
+        event_type: str = "Warning",
+    ) -> None:
+        """Emit a core/v1 Event on a pod (binding failures etc.). Best-effort:
+        errors are swallowed by callers — events must never block binding."""
+        import datetime
+
+        now = datetime.datetime.now(datetime.timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
+        body = {
+            "metadata": {"generateName": "egpu-", "namespace": namespace},
+            "involvedObject": {
+                "kind": "Pod",
+                "namespace": namespace,
+                "name": pod_name,
+                "uid": pod_uid,
+            },
+            "reason": reason,
+            "message": message[:1024],
+            "type": event_type,
+            "source": {"component": "elastic-gpu-agent-amd"},
+            "firstTimestamp": now,
+            "lastTimestamp": now,
+            "count": 1,
+        }
+        r = self._client.post(f"/api/v1/namespaces/{namespace}/events", json=body)
+        r.raise_for_status()
+
     def get_pod(self, namespace: str, name: str) -> Pod:
         r = self._client.get(f"/api/v1/namespaces/{namespace}/pods/{name}")
         if r.status_code == 404:

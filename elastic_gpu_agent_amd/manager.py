@@ -66,6 +66,7 @@ class GPUManager:
 
             client = K8sClient(kubeconf=opts.kubeconf)
             self.sitter = PodSitter(client, opts.node_name, delete_hook=self._on_pod_delete)
+            self._event_client = client
 
         if locators is not None:
             core_loc, mem_loc = locators
@@ -80,6 +81,17 @@ class GPUManager:
         limits = LimitsWriter(opts.paths.limits_dir)
         cumask = CUMaskAllocator(self.storage, self.operator.devices())
 
+        event_sink = None
+        client = getattr(self, "_event_client", None)
+        if client is not None:
+            def event_sink(ns, pod, reason, message, _c=client):
+                try:
+                    p = self.sitter.get_pod(ns, pod)
+                    uid = p.uid
+                except Exception:
+                    uid = ""
+                _c.create_event(ns, pod, uid, reason, message)
+
         self.config = GPUPluginConfig(
             operator=self.operator,
             storage=self.storage,
@@ -90,6 +102,7 @@ class GPUManager:
             options=opts.plugin_options,
             limits=limits,
             cumask=cumask,
+            event_sink=event_sink,
         )
         self.plugin = GPUSharePlugin(self.config)
         self._gc_thread: Optional[threading.Thread] = None

@@ -42,3 +42,6 @@ class GPUPluginConfig:
     options: PluginOptions = dataclasses.field(default_factory=PluginOptions)
     limits: Optional[object] = None  # isolation.LimitsWriter
     cumask: Optional[object] = None  # isolation.CUMaskAllocator
+    # optional callable(namespace, pod_name, reason, message) emitting a k8s
+    # Event; must be non-blocking best-effort
+    event_sink: Optional[object] = None

@@ -194,6 +194,7 @@ class GPUSharePluginBase:
         except Exception as e:
             for alloc_id in created:  # rollback partial symlinks
                 self.cfg.operator.delete(-1, alloc_id)
+            self._emit_event(pc, "EgpuBindFailed", f"bind {device.hash}: {e}")
             return self._fail(context, f"bind {device.hash}: {e}")
         pi = self.cfg.storage.load_or_create(pc.namespace, pc.name)
         pi.container_device_map[pc.container] = device
@@ -205,6 +206,15 @@ class GPUSharePluginBase:
 
     def _locator(self):
         raise NotImplementedError
+
+    def _emit_event(self, pc, reason, message):
+        sink = self.cfg.event_sink
+        if sink is None:
+            return
+        try:
+            sink(pc.namespace, pc.name, reason, message)
+        except Exception as e:  # events are best-effort
+            log.debug("event emission failed: %s", e)
 
     @staticmethod
     def _fail(context, msg):

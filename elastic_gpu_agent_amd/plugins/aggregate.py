@@ -107,6 +107,11 @@ class GPUSharePlugin:
         while not self._stop.is_set():
             try:
                 gc_events.get(timeout=period)
+                while True:  # coalesce bursts of delete events into one pass
+                    try:
+                        gc_events.get_nowait()
+                    except queue.Empty:
+                        break
             except queue.Empty:
                 pass
             try:

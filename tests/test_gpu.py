@@ -300,6 +300,42 @@ def test_hook_real_injection(tmp_path, gpus):
         target.wait()
 
 
+def test_occupancy_report_on_gpu(tmp_path, gpus):
+    """egpuctl occupancy path against the real amdsmi: per-GPU telemetry is
+    present and pod rows join with mask metadata."""
+    sys.path.insert(0, os.path.join(REPO, "tests"))
+    from helpers import Harness
+
+    from elastic_gpu_agent_amd import consts
+    from elastic_gpu_agent_amd.isolation import CUMaskAllocator
+    from elastic_gpu_agent_amd.isolation.occupancy import report
+    from elastic_gpu_agent_amd.operator import GPUOperator
+    from elastic_gpu_agent_amd.operator.amdsmi import AmdSmiBackend
+    from elastic_gpu_agent_amd.types import Device, PodContainer
+
+    h = Harness(str(tmp_path), gpus=1)
+    backend = AmdSmiBackend()
+    h.plugin.cfg.operator = GPUOperator(backend, dev_root=h.paths.dev_root)
+    h.plugin.cfg.cumask = CUMaskAllocator(h.storage, backend.devices())
+    g0 = backend.devices()[0]
+    ids = [f"{g0.index}-{i:02d}" for i in range(25)]
+    d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    h.core_locator.assign(d.hash, PodContainer("ns", "occ", "main"))
+    h.add_assumed_pod("ns", "occ", "main", str(g0.index))
+    h.plugin.core.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    h.plugin.core.pre_start_container({"devicesIDs": ids}, None)
+
+    from elastic_gpu_agent_amd import _amdsmi
+
+    rep = report(h.storage, h.plugin.cfg.limits, state_dir=str(tmp_path), smi=_amdsmi)
+    assert rep["pods"][0]["pod"] == "ns/occ"
+    assert rep["pods"][0]["cu_limit"] == 64
+    telemetry = rep["gpus"][g0.index]
+    assert "error" not in telemetry
+    assert telemetry.get("vram_total_mb", 0) > 100_000  # 288 GB part
+    h.close()
+
+
 def test_shim_blocks_mask_widening(gpus):
     """A container calling cu_set_mask itself cannot widen past its quota:
     the shim intersects requests with the allocation mask."""

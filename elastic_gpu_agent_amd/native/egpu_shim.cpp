@@ -27,12 +27,14 @@
 #include <hsa/hsa_ext_amd.h>
 
 #include <atomic>
+#include <csignal>
 #include <cstdarg>
 #include <cstdint>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
 #include <dirent.h>
+#include <execinfo.h>
 #include <mutex>
 #include <string>
 #include <unordered_map>
@@ -129,9 +131,28 @@ std::string slurp(const std::string& path) {
   return out;
 }
 
+// deny status for over-quota allocations (default OUT_OF_RESOURCES; the
+// EGPU_DENY_STATUS knob exists for runtime-compat experiments)
+hsa_status_t g_deny_status = HSA_STATUS_ERROR_OUT_OF_RESOURCES;
+
+void segv_handler(int sig) {
+  void* frames[64];
+  int n = backtrace(frames, 64);
+  fprintf(stderr, "[egpu-shim] FATAL signal %d; backtrace (%d frames):\n", sig, n);
+  backtrace_symbols_fd(frames, n, 2);
+  signal(sig, SIG_DFL);
+  raise(sig);
+}
+
 void load_config() {
   const char* verbose = getenv("EGPU_SHIM_VERBOSE");
   g_verbose = verbose && verbose[0] == '1';
+  if (const char* dbg = getenv("EGPU_SHIM_DEBUG"); dbg && dbg[0] == '1') {
+    signal(SIGSEGV, segv_handler);
+    signal(SIGABRT, segv_handler);
+  }
+  if (const char* ds = getenv("EGPU_DENY_STATUS"))
+    g_deny_status = static_cast<hsa_status_t>(atoi(ds));
 
   const char* dir = getenv("EGPU_LIMITS_DIR");
   std::string limits_dir = dir ? dir : "/etc/egpu";
@@ -296,7 +317,7 @@ hsa_status_t pool_allocate_wrap(hsa_amd_memory_pool_t pool, size_t size, uint32_
       logf("DENY pool alloc %zu (used %llu / limit %llu)", size, (unsigned long long)prev,
            (unsigned long long)g_cfg.mem_limit);
       if (ptr) *ptr = nullptr;  // some callers check the pointer, not status
-      return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
+      return g_deny_status;
     }
     hsa_status_t st = g_amdext.hsa_amd_memory_pool_allocate_fn(pool, size, flags, ptr);
     if (st != HSA_STATUS_SUCCESS) {
@@ -321,7 +342,7 @@ hsa_status_t memory_allocate_wrap(hsa_region_t region, size_t size, void** ptr) 
       g_vram_used.fetch_sub(size);
       g_denied.fetch_add(1);
       if (ptr) *ptr = nullptr;
-      return HSA_STATUS_ERROR_OUT_OF_RESOURCES;
+      return g_deny_status;
     }
     hsa_status_t st = g_core.hsa_memory_allocate_fn(region, size, ptr);
     if (st != HSA_STATUS_SUCCESS) {

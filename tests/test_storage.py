@@ -152,3 +152,32 @@ def test_is_bolt_file_rejects_sqlite(tmp_db):
     st2 = new_storage(tmp_db)
     assert st2.load("ns", "p1").name == "p1"
     st2.close()
+
+
+def test_concurrent_access(tmp_db):
+    """Threaded save/load/delete hammering — the store must stay consistent."""
+    import threading
+
+    st = Storage(tmp_db)
+    errors = []
+
+    def writer(tid):
+        try:
+            for i in range(50):
+                st.save(make_pi(name=f"t{tid}-p{i % 5}", ids=(f"0-{i:02d}",)))
+                if i % 7 == 0:
+                    st.delete("ns", f"t{tid}-p{i % 5}")
+                st.for_each(lambda pi: None)
+                st.aux_set(f"k{tid}", str(i))
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=writer, args=(t,)) for t in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors
+    # final state readable and parseable
+    st.for_each(lambda pi: pi.val())
+    st.close()

@@ -66,6 +66,7 @@ def test_census_unmasked(gpus):
 def _run_masked(pyexpr: str, extra_env: dict) -> str:
     env = dict(os.environ)
     env["HSA_TOOLS_LIB"] = SHIM
+    env["EGPU_SHIM_DEBUG"] = "1"  # backtrace on any crash in the shimmed run
     env.update(extra_env)
     out = subprocess.run(
         [sys.executable, "-c", pyexpr], env=env, cwd=REPO,

@@ -16,6 +16,8 @@ GPU_PERCENT_EACH_CARD = 100
 # Pod annotations written by the elastic-gpu-scheduler and consumed here.
 ELASTIC_GPU_ASSUMED_ANNOTATION = "elasticgpu.io/assumed"
 ELASTIC_GPU_CONTAINER_ANNOTATION = "elasticgpu.io/container-%s"
+# optional per-pod QoS class consumed by the HSA shim (queue priority):
+ELASTIC_GPU_QOS_ANNOTATION = "elasticgpu.io/qos-class"  # low | normal | high
 
 NODE_NAME_FIELD = "spec.nodeName"
 

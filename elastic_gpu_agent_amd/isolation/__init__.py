@@ -167,6 +167,7 @@ class LimitsWriter:
         cu_mask: Optional[str] = None,
         cu_count: Optional[int] = None,
         mem_limit_bytes: Optional[int] = None,
+        priority: Optional[str] = None,
     ) -> None:
         dev_by_idx = {d.index: d for d in devices}
         rec: Dict = {
@@ -182,6 +183,8 @@ class LimitsWriter:
             rec["cu_count"] = cu_count
         if mem_limit_bytes is not None:
             rec["mem_limit_bytes"] = mem_limit_bytes
+        if priority is not None:
+            rec["priority"] = priority
         self._atomic_write(self.host_path(alloc_hash), rec)
 
     def delete(self, alloc_hash: str) -> None:

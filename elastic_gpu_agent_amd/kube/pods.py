@@ -35,6 +35,10 @@ class Pod:
         (annotation contract, ref: pkg/common/const.go:7)."""
         return self.annotations.get(consts.ELASTIC_GPU_ASSUMED_ANNOTATION) == "true"
 
+    def qos_class(self) -> Optional[str]:
+        v = self.annotations.get(consts.ELASTIC_GPU_QOS_ANNOTATION)
+        return v if v in ("low", "normal", "high") else None
+
     def container_gpu_indexes(self, container: str) -> Optional[str]:
         """Raw value of the per-container binding annotation
         (comma-separated GPU indexes, ref: pkg/plugins/gpushare.go:107-125)."""

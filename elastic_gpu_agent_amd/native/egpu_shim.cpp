@@ -64,6 +64,7 @@ struct Config {
   uint32_t mask[kMaskWordsMax] = {0};
   int mask_words = 0;
   uint64_t mem_limit = 0;  // 0 = unlimited
+  int priority = -1;       // -1 unset; 0 low / 1 normal / 2 high (QoS class)
 };
 
 Config g_cfg;
@@ -170,6 +171,12 @@ void load_config() {
       if (json_find_string(body, "cu_mask", &mask)) parse_mask_hex(mask.c_str(), &g_cfg);
       uint64_t mem = 0;
       if (json_find_u64(body, "mem_limit_bytes", &mem) && mem > 0) g_cfg.mem_limit = mem;
+      std::string prio;
+      if (json_find_string(body, "priority", &prio)) {
+        if (prio == "low") g_cfg.priority = 0;
+        else if (prio == "normal") g_cfg.priority = 1;
+        else if (prio == "high") g_cfg.priority = 2;
+      }
       logf("loaded %s (mask=%d mem_limit=%llu)", name.c_str(), g_cfg.have_mask ? 1 : 0,
            (unsigned long long)g_cfg.mem_limit);
     }
@@ -178,6 +185,7 @@ void load_config() {
   // env overrides (tests / manual runs)
   if (const char* m = getenv("EGPU_CU_MASK")) parse_mask_hex(m, &g_cfg);
   if (const char* l = getenv("EGPU_MEM_LIMIT_BYTES")) g_cfg.mem_limit = strtoull(l, nullptr, 10);
+  if (const char* pr = getenv("EGPU_PRIORITY")) g_cfg.priority = atoi(pr);
   logf("config: have_mask=%d words=%d mem_limit=%llu", g_cfg.have_mask, g_cfg.mask_words,
        (unsigned long long)g_cfg.mem_limit);
 }
@@ -285,7 +293,17 @@ hsa_status_t queue_create_wrap(hsa_agent_t agent, uint32_t size, hsa_queue_type3
                                hsa_queue_t** queue) {
   hsa_status_t st = g_core.hsa_queue_create_fn(agent, size, type, callback, data,
                                                private_segment_size, group_segment_size, queue);
-  if (st == HSA_STATUS_SUCCESS && queue && *queue) apply_mask(*queue, agent);
+  if (st == HSA_STATUS_SUCCESS && queue && *queue) {
+    apply_mask(*queue, agent);
+    if (g_cfg.priority >= 0) {
+      static const hsa_amd_queue_priority_t prios[3] = {
+          HSA_AMD_QUEUE_PRIORITY_LOW, HSA_AMD_QUEUE_PRIORITY_NORMAL,
+          HSA_AMD_QUEUE_PRIORITY_HIGH};
+      int p = g_cfg.priority > 2 ? 2 : g_cfg.priority;
+      hsa_status_t pst = g_amdext.hsa_amd_queue_set_priority_fn(*queue, prios[p]);
+      if (pst == HSA_STATUS_SUCCESS) logf("queue %p priority -> %d", (void*)*queue, p);
+    }
+  }
   return st;
 }
 

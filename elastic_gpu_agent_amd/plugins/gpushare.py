@@ -190,7 +190,7 @@ class GPUSharePluginBase:
             return self._fail(context, f"bad GPU index annotation {raw!r}")
         created: List[str] = []
         try:
-            self._bind(device, ids, indexes, created)
+            self._bind(device, ids, indexes, created, pod)
         except Exception as e:
             for alloc_id in created:  # rollback partial symlinks
                 self.cfg.operator.delete(-1, alloc_id)
@@ -201,7 +201,8 @@ class GPUSharePluginBase:
         self.cfg.storage.save(pi)
         return {}
 
-    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str]):
+    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str],
+              pod=None):
         raise NotImplementedError
 
     def _locator(self):
@@ -265,7 +266,8 @@ class GPUShareCorePlugin(GPUSharePluginBase):
             resp["mounts"] = iso["mounts"]
         return resp
 
-    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str]):
+    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str],
+              pod=None):
         n_links = self.links_for(len(ids))
         if len(indexes) != n_links:
             raise ValueError(
@@ -276,6 +278,7 @@ class GPUShareCorePlugin(GPUSharePluginBase):
             self.cfg.operator.create(idx, alloc_id)
             created.append(alloc_id)
         percent = len(ids)
+        priority = pod.qos_class() if pod is not None else None
         if percent < consts.GPU_PERCENT_EACH_CARD and self.cfg.cumask and self.cfg.limits:
             mask_hex, n_cus = self.cfg.cumask.allocate(device.hash, indexes[0], percent)
             self.cfg.limits.finalize(
@@ -284,10 +287,12 @@ class GPUShareCorePlugin(GPUSharePluginBase):
                 devices=self.cfg.operator.devices(),
                 cu_mask=mask_hex,
                 cu_count=n_cus,
+                priority=priority,
             )
         elif self.cfg.limits:
             self.cfg.limits.finalize(
-                device.hash, gpu_indexes=indexes, devices=self.cfg.operator.devices()
+                device.hash, gpu_indexes=indexes, devices=self.cfg.operator.devices(),
+                priority=priority,
             )
 
     def _locator(self):
@@ -324,7 +329,8 @@ class GPUShareMemoryPlugin(GPUSharePluginBase):
             resp["mounts"] = iso["mounts"]
         return resp
 
-    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str]):
+    def _bind(self, device: Device, ids: List[str], indexes: List[int], created: List[str],
+              pod=None):
         if len(indexes) != 1:
             raise ValueError(f"memory binding expects exactly 1 GPU index, got {indexes}")
         alloc_id = f"{device.hash}-0"
@@ -337,6 +343,7 @@ class GPUShareMemoryPlugin(GPUSharePluginBase):
                 gpu_indexes=indexes,
                 devices=self.cfg.operator.devices(),
                 mem_limit_bytes=mem_bytes,
+                priority=pod.qos_class() if pod is not None else None,
             )
 
     def _locator(self):

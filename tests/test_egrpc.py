@@ -258,5 +258,8 @@ def test_latency_beats_grpcio_floor(eserver):
         lat.append(time.perf_counter() - t0)
     lat.sort()
     p50 = lat[len(lat) // 2] * 1e6
-    assert p50 < 500, f"egrpc p50 {p50:.0f}µs — transport regression"
+    # grpcio's floor on this host class is ~1200 µs; egrpc runs ~100-300 µs.
+    # The bound is intentionally loose — CI containers are noisy — but still
+    # below anything grpcio can do, so a transport regression trips it.
+    assert p50 < 900, f"egrpc p50 {p50:.0f}µs — transport regression"
     ch.close()

@@ -199,7 +199,19 @@ class LimitsWriter:
 
     @staticmethod
     def _atomic_write(path: str, obj: dict) -> None:
-        tmp = path + ".tmp"
-        with open(tmp, "w") as f:
-            json.dump(obj, f)
-        os.replace(tmp, path)
+        # unique temp name: concurrent writers of the SAME allocation (e.g. a
+        # kubelet retry racing the first attempt) must not clobber each
+        # other's temp file (found by tools/soak.py)
+        import tempfile
+
+        fd, tmp = tempfile.mkstemp(dir=os.path.dirname(path), suffix=".tmp")
+        try:
+            with os.fdopen(fd, "w") as f:
+                json.dump(obj, f)
+            os.replace(tmp, path)
+        except BaseException:
+            try:
+                os.unlink(tmp)
+            except OSError:
+                pass
+            raise

@@ -68,6 +68,24 @@ def build_fastwire(force=False):
     return out
 
 
+def build_etransport(force=False):
+    src = os.path.join(HERE, "etransport.cpp")
+    tables = os.path.join(HERE, "hpack_tables.h")
+    gen = os.path.join(HERE, "gen_hpack_tables.py")
+    if not os.path.exists(tables) or os.path.getmtime(tables) < os.path.getmtime(gen):
+        _run([sys.executable, gen])
+    ext = sysconfig.get_config_var("EXT_SUFFIX")
+    out = os.path.join(PKG, f"_etransport{ext}")
+    if not force and _newer(out, src, tables):
+        return out
+    _run(
+        ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src]
+        + pybind_includes()
+        + ["-pthread", "-o", out]
+    )
+    return out
+
+
 def build_shim(force=False):
     src = os.path.join(HERE, "egpu_shim.cpp")
     out = os.path.join(PKG, "libegpu_shim.so")
@@ -107,6 +125,7 @@ def build_all(force=False):
     return {
         "amdsmi": build_amdsmi(force),
         "fastwire": build_fastwire(force),
+        "etransport": build_etransport(force),
         "shim": build_shim(force),
         "kernels": build_kernels(force),
         "hook": build_hook(force),

@@ -14,10 +14,21 @@ from typing import Iterator, List, Optional, Tuple
 from . import core, hpack
 from .core import ConnectionClosed, EgrpcError, frame_header, parse_frame_header
 
+try:
+    from elastic_gpu_agent_amd import _etransport
+except ImportError:  # pure-Python fallback (also forced by EGPU_PY_TRANSPORT=1)
+    _etransport = None
+
+
+def _use_native() -> bool:
+    import os
+
+    return _etransport is not None and os.environ.get("EGPU_PY_TRANSPORT") != "1"
+
 
 class Channel:
     def __init__(self, unix_path: str, connect_timeout: float = 10.0,
-                 authority: bytes = b"localhost"):
+                 authority: bytes = b"localhost", _force_python: bool = False):
         self._path = unix_path
         self._authority = authority
         self._lock = threading.RLock()
@@ -33,6 +44,12 @@ class Channel:
         self._stream_recv_deficit = 0
         self._deficit_sid = 0
         self._active_stream_window = core.DEFAULT_WINDOW
+        # C++ unary data plane (native/etransport.cpp); streaming calls
+        # delegate to a lazily-created pure-Python sibling channel
+        self._native = None
+        self._stream_channel: Optional["Channel"] = None
+        if not _force_python and _use_native():
+            self._native = _etransport.ClientCore(unix_path)
 
     # ---- connection ----
     def _connect(self) -> None:
@@ -70,6 +87,11 @@ class Channel:
 
     def close(self) -> None:
         with self._lock:
+            if self._native is not None:
+                self._native.close()
+            if self._stream_channel is not None:
+                self._stream_channel.close()
+                self._stream_channel = None
             if self._sock is not None:
                 try:
                     self._sock.close()
@@ -223,8 +245,44 @@ class Channel:
                 raise EgrpcError(core.UNAVAILABLE, "stream reset")
 
     # ---- public API ----
+    def _native_connect(self):
+        deadline = time.time() + self._connect_timeout
+        last = None
+        while time.time() < deadline:
+            try:
+                self._native.connect()
+                return
+            except RuntimeError as e:
+                last = e
+                time.sleep(0.05)
+        raise EgrpcError(core.UNAVAILABLE, f"connect {self._path}: {last}")
+
     def unary_unary(self, path: str, request_serializer=None, response_deserializer=None):
         pbytes = path.encode()
+        if self._native is not None:
+            header_block = self._request_headers(pbytes)
+            native = self._native
+
+            def native_call(request, timeout: Optional[float] = None):
+                msg = request_serializer(request) if request_serializer else request
+                with self._lock:
+                    if not native.connected():
+                        self._native_connect()
+                    try:
+                        status, data, gmsg = native.call_unary(
+                            header_block, msg, timeout or 0.0
+                        )
+                    except RuntimeError as e:
+                        native.close()
+                        if "timeout" in str(e):
+                            raise EgrpcError(core.DEADLINE_EXCEEDED, path) from e
+                        raise EgrpcError(core.UNAVAILABLE,
+                                         f"connection lost: {e}") from e
+                if status != 0:
+                    raise EgrpcError(status, gmsg)
+                return response_deserializer(data) if response_deserializer else data
+
+            return native_call
         # the header block is constant per method (stateless encoder):
         # precompute it once instead of per call
         header_block = self._request_headers(pbytes)
@@ -253,6 +311,15 @@ class Channel:
         return call
 
     def unary_stream(self, path: str, request_serializer=None, response_deserializer=None):
+        if self._native is not None:
+            if self._stream_channel is None:
+                self._stream_channel = Channel(
+                    self._path, self._connect_timeout, self._authority,
+                    _force_python=True,
+                )
+            return self._stream_channel.unary_stream(
+                path, request_serializer, response_deserializer
+            )
         pbytes = path.encode()
 
         def call(request, timeout: Optional[float] = None) -> Iterator:

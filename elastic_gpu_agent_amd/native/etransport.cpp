@@ -23,6 +23,8 @@
 #include <memory>
 #include <mutex>
 #include <string>
+#include <csignal>
+#include <execinfo.h>
 #include <sys/socket.h>
 #include <sys/un.h>
 #include <thread>
@@ -303,13 +305,22 @@ struct CoreShared {
 };
 
 // ---------------------------------------------------------------- connection
-class Connection {
+class Connection : public std::enable_shared_from_this<Connection> {
  public:
   Connection(int fd, std::shared_ptr<CoreShared> core)
       : fd_(fd), core_(std::move(core)) {}
 
   void run() {
-    if (!expect_preface()) { finish(); return; }
+    try {
+      run_inner();
+    } catch (...) {
+      // defensive: no exception may escape a detached thread
+    }
+    finish();
+  }
+
+  void run_inner() {
+    if (!expect_preface()) { return; }
     {
       // our SETTINGS + connection window grant
       uint8_t out[9 + 18 + 13];
@@ -343,7 +354,6 @@ class Connection {
         if (stp) dispatch(*stp);
       }
     }
-    finish();
   }
 
   void close_now() {
@@ -714,9 +724,15 @@ class Connection {
         pack = new Pack{it->second.fn, core_->context_factory(), std::move(request)};
         st.ctx = pack->ctx;
       }
-      Connection* self = this;
+      // the thread must keep the Connection alive: a raw `this` would dangle
+      // once the client disconnects and the server prunes the connection
+      std::shared_ptr<Connection> self = shared_from_this();
       std::thread([self, pack, sid]() {
-        self->run_streaming(pack->fn, pack->ctx, pack->request, sid);
+        try {
+          self->run_streaming(pack->fn, pack->ctx, pack->request, sid);
+        } catch (...) {
+          self->close_now();
+        }
         if (Py_IsInitialized()) {
           py::gil_scoped_acquire gil;
           delete pack;
@@ -982,7 +998,15 @@ class ServerCore {
               conns_.end());
         }
       }
-      std::thread([conn]() { conn->run(); }).detach();
+      std::thread([conn]() {
+        try {
+          conn->run();
+        } catch (...) {
+          // a stray exception in a detached thread would std::terminate the
+          // whole process; drop the connection instead
+          conn->close_now();
+        }
+      }).detach();
     }
   }
 
@@ -994,11 +1018,341 @@ class ServerCore {
   std::vector<std::shared_ptr<Connection>> conns_;
 };
 
+
+// ---------------------------------------------------------------- client
+// Blocking unary client connection. Python owns reconnect policy; this core
+// owns one connected socket and performs whole unary calls with the GIL
+// released. Streaming calls stay on the Python client implementation.
+class ClientCore {
+ public:
+  explicit ClientCore(const std::string& path) : path_(path) {}
+
+  ~ClientCore() { close_fd(); }
+
+  void connect() {
+    close_fd();
+    fd_ = ::socket(AF_UNIX, SOCK_STREAM, 0);
+    if (fd_ < 0) throw std::runtime_error("socket() failed");
+    sockaddr_un addr{};
+    addr.sun_family = AF_UNIX;
+    if (path_.size() >= sizeof(addr.sun_path)) {
+      close_fd();
+      throw std::runtime_error("socket path too long");
+    }
+    strcpy(addr.sun_path, path_.c_str());
+    if (::connect(fd_, (sockaddr*)&addr, sizeof(addr)) != 0) {
+      close_fd();
+      throw std::runtime_error("connect failed: " + path_);
+    }
+    buf_.clear();
+    pos_ = 0;
+    next_stream_ = 1;
+    decoder_ = HpackDecoder();
+    peer_max_frame_ = 16384;
+    peer_initial_window_ = DEFAULT_WINDOW;
+    conn_send_window_ = DEFAULT_WINDOW;
+    conn_recv_deficit_ = 0;
+    // preface + SETTINGS + window grant
+    std::string out((const char*)"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n", 24);
+    uint8_t f[9 + 12 + 13];
+    uint8_t* p = f;
+    put_frame_header(p, 12, F_SETTINGS, 0, 0); p += 9;
+    auto put_setting = [&](uint16_t k, uint32_t v) {
+      p[0] = k >> 8; p[1] = k; uint32_t nv = htonl(v); memcpy(p + 2, &nv, 4); p += 6;
+    };
+    put_setting(S_MAX_FRAME, OUR_MAX_FRAME);
+    put_setting(S_INITIAL_WINDOW, (uint32_t)RECV_WINDOW);
+    put_frame_header(p, 4, F_WINUP, 0, 0);
+    uint32_t inc = htonl((uint32_t)(RECV_WINDOW - DEFAULT_WINDOW));
+    memcpy(p + 9, &inc, 4);
+    p += 13;
+    out.append((const char*)f, p - f);
+    write_all(out.data(), out.size());
+  }
+
+  bool connected() const { return fd_ >= 0; }
+
+  void close_fd() {
+    if (fd_ >= 0) { ::close(fd_); fd_ = -1; }
+  }
+
+  // returns (grpc_status, response_bytes, grpc_message). Transport failures
+  // throw std::runtime_error (Python resets + maps to UNAVAILABLE).
+  py::tuple call_unary(py::bytes header_block_b, py::bytes message_b, double timeout_s) {
+    std::string header_block = header_block_b;
+    std::string message = message_b;
+    int status = 0;
+    std::string data, grpc_message, err;
+    {
+      py::gil_scoped_release release;
+      try {
+        do_call(header_block, message, timeout_s, &status, &data, &grpc_message);
+      } catch (const std::exception& e) {
+        err = e.what();
+      }
+    }
+    if (!err.empty()) throw std::runtime_error(err);
+    return py::make_tuple(status, py::bytes(data), grpc_message);
+  }
+
+ private:
+  void set_timeout(double seconds) {
+    timeval tv{};
+    if (seconds > 0) {
+      tv.tv_sec = (time_t)seconds;
+      tv.tv_usec = (suseconds_t)((seconds - tv.tv_sec) * 1e6);
+    }
+    setsockopt(fd_, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+  }
+
+  void write_all(const void* data, size_t n) {
+    const uint8_t* p = (const uint8_t*)data;
+    size_t off = 0;
+    while (off < n) {
+      ssize_t w = ::write(fd_, p + off, n - off);
+      if (w <= 0) throw std::runtime_error("connection lost (write)");
+      off += w;
+    }
+  }
+
+  bool fill(size_t need) {
+    while (buf_.size() - pos_ < need) {
+      uint8_t tmp[262144];
+      ssize_t r = ::read(fd_, tmp, sizeof(tmp));
+      if (r == 0) throw std::runtime_error("connection closed");
+      if (r < 0) {
+        if (errno == EAGAIN || errno == EWOULDBLOCK)
+          throw std::runtime_error("timeout");
+        throw std::runtime_error("connection lost (read)");
+      }
+      buf_.append((const char*)tmp, r);
+      if (pos_ > (1 << 20) && pos_ * 2 > buf_.size()) {
+        buf_.erase(0, pos_);
+        pos_ = 0;
+      }
+    }
+    return true;
+  }
+
+  void read_frame(uint8_t* type, uint8_t* flags, uint32_t* sid, std::string* body) {
+    fill(9);
+    const uint8_t* h = (const uint8_t*)buf_.data() + pos_;
+    uint32_t len = h[0] << 16 | h[1] << 8 | h[2];
+    *type = h[3];
+    *flags = h[4];
+    uint32_t sraw;
+    memcpy(&sraw, h + 5, 4);
+    *sid = ntohl(sraw) & 0x7fffffff;
+    if (len > 64u * 1024 * 1024) throw std::runtime_error("oversize frame");
+    fill(9 + len);
+    body->assign(buf_.data() + pos_ + 9, len);
+    pos_ += 9 + len;
+  }
+
+  // conn-level frames; returns true if consumed
+  bool handle_conn_frame(uint8_t type, uint8_t flags, uint32_t sid,
+                         const std::string& body, int64_t* active_stream_window) {
+    if (type == F_SETTINGS) {
+      if (!(flags & FLAG_ACK)) {
+        for (size_t off = 0; off + 6 <= body.size(); off += 6) {
+          uint16_t k = (uint8_t)body[off] << 8 | (uint8_t)body[off + 1];
+          uint32_t v;
+          memcpy(&v, body.data() + off + 2, 4);
+          v = ntohl(v);
+          if (k == S_MAX_FRAME) peer_max_frame_ = v;
+          else if (k == S_INITIAL_WINDOW) {
+            int64_t delta = (int64_t)v - peer_initial_window_;
+            peer_initial_window_ = v;
+            *active_stream_window += delta;  // RFC 7540 §6.9.2
+          }
+        }
+        uint8_t ack[9];
+        put_frame_header(ack, 0, F_SETTINGS, FLAG_ACK, 0);
+        write_all(ack, 9);
+      }
+      return true;
+    }
+    if (type == F_PING) {
+      if (!(flags & FLAG_ACK) && body.size() == 8) {
+        uint8_t p[17];
+        put_frame_header(p, 8, F_PING, FLAG_ACK, 0);
+        memcpy(p + 9, body.data(), 8);
+        write_all(p, 17);
+      }
+      return true;
+    }
+    if (type == F_WINUP && sid == 0) {
+      uint32_t inc;
+      memcpy(&inc, body.data(), 4);
+      conn_send_window_ += ntohl(inc) & 0x7fffffff;
+      return true;
+    }
+    if (type == F_GOAWAY) throw std::runtime_error("server sent GOAWAY");
+    return false;
+  }
+
+  void do_call(const std::string& header_block, const std::string& message,
+               double timeout_s, int* status, std::string* data,
+               std::string* grpc_message) {
+    set_timeout(timeout_s);
+    uint32_t sid = next_stream_;
+    next_stream_ += 2;
+    int64_t stream_window = peer_initial_window_;
+
+    std::string payload;
+    payload.resize(5 + message.size());
+    payload[0] = 0;
+    uint32_t mlen = htonl((uint32_t)message.size());
+    memcpy(&payload[1], &mlen, 4);
+    memcpy(&payload[5], message.data(), message.size());
+
+    std::string out;
+    out.resize(9);
+    put_frame_header((uint8_t*)out.data(), header_block.size(), F_HEADERS,
+                     FLAG_END_HEADERS, sid);
+    out += header_block;
+    size_t off = 0, total = payload.size();
+    while (true) {
+      int64_t avail = std::min(conn_send_window_, stream_window);
+      if ((int64_t)peer_max_frame_ < avail) avail = peer_max_frame_;
+      if (total - off > 0 && avail <= 0) {
+        write_all(out.data(), out.size());
+        out.clear();
+        uint8_t type, flags; uint32_t fsid; std::string body;
+        read_frame(&type, &flags, &fsid, &body);
+        if (!handle_conn_frame(type, flags, fsid, body, &stream_window)) {
+          if (type == F_WINUP && fsid == sid) {
+            uint32_t inc;
+            memcpy(&inc, body.data(), 4);
+            stream_window += ntohl(inc) & 0x7fffffff;
+          } else if (type == F_RST) {
+            throw std::runtime_error("stream reset during send");
+          }
+        }
+        continue;
+      }
+      size_t n = std::min(total - off, (size_t)avail);
+      bool last = off + n >= total;
+      uint8_t flags = last ? FLAG_END_STREAM : 0;
+      size_t fo = out.size();
+      out.resize(fo + 9 + n);
+      put_frame_header((uint8_t*)out.data() + fo, n, F_DATA, flags, sid);
+      memcpy(&out[fo + 9], payload.data() + off, n);
+      conn_send_window_ -= n;
+      stream_window -= n;
+      off += n;
+      if (last) break;
+    }
+    write_all(out.data(), out.size());
+
+    // response
+    std::string rdata;
+    std::vector<std::pair<std::string, std::string>> headers;
+    while (true) {
+      uint8_t type, flags; uint32_t fsid; std::string body;
+      read_frame(&type, &flags, &fsid, &body);
+      if (handle_conn_frame(type, flags, fsid, body, &stream_window)) continue;
+      if (fsid != sid) continue;
+      if (type == F_HEADERS || type == F_CONT) {
+        size_t hoff = 0, pad = 0;
+        if (type == F_HEADERS && (flags & FLAG_PADDED)) { pad = (uint8_t)body[0]; hoff = 1; }
+        if (type == F_HEADERS && (flags & FLAG_PRIORITY)) hoff += 5;
+        if (!decoder_.decode((const uint8_t*)body.data() + hoff,
+                             body.size() - hoff - pad, &headers))
+          throw std::runtime_error("bad hpack from server");
+        if (flags & FLAG_END_STREAM) {
+          int st = 0;
+          std::string msg;
+          for (auto& h : headers) {
+            if (h.first == "grpc-status") st = atoi(h.second.c_str());
+            else if (h.first == "grpc-message") msg = h.second;
+          }
+          // grpc-message is percent-encoded; decode minimally
+          std::string dec;
+          for (size_t i = 0; i < msg.size(); ++i) {
+            if (msg[i] == '%' && i + 2 < msg.size()) {
+              dec.push_back((char)strtol(msg.substr(i + 1, 2).c_str(), nullptr, 16));
+              i += 2;
+            } else dec.push_back(msg[i]);
+          }
+          // parse first grpc frame of rdata
+          std::string first;
+          if (rdata.size() >= 5) {
+            uint32_t len;
+            memcpy(&len, rdata.data() + 1, 4);
+            len = ntohl(len);
+            if (5 + (size_t)len <= rdata.size()) first = rdata.substr(5, len);
+          }
+          *status = st;
+          *data = first;
+          *grpc_message = dec;
+          return;
+        }
+      } else if (type == F_DATA) {
+        rdata += body;
+        if (!body.empty()) {
+          conn_recv_deficit_ += body.size();
+          if (conn_recv_deficit_ >= RECV_WINDOW / 2) {
+            uint8_t f2[26];
+            put_frame_header(f2, 4, F_WINUP, 0, 0);
+            uint32_t v = htonl((uint32_t)conn_recv_deficit_);
+            memcpy(f2 + 9, &v, 4);
+            put_frame_header(f2 + 13, 4, F_WINUP, 0, sid);
+            memcpy(f2 + 22, &v, 4);
+            write_all(f2, 26);
+            conn_recv_deficit_ = 0;
+          }
+        }
+        if (flags & FLAG_END_STREAM)
+          throw std::runtime_error("stream ended without trailers");
+      } else if (type == F_RST) {
+        throw std::runtime_error("stream reset");
+      }
+    }
+  }
+
+  std::string path_;
+  int fd_ = -1;
+  std::string buf_;
+  size_t pos_ = 0;
+  uint32_t next_stream_ = 1;
+  HpackDecoder decoder_;
+  uint32_t peer_max_frame_ = 16384;
+  int64_t peer_initial_window_ = DEFAULT_WINDOW;
+  int64_t conn_send_window_ = DEFAULT_WINDOW;
+  int64_t conn_recv_deficit_ = 0;
+};
+
 }  // namespace
 
 PYBIND11_MODULE(_etransport, m) {
   trie_init();
+  if (const char* dbg = getenv("EGPU_ETRANSPORT_DEBUG"); dbg && dbg[0] == '1') {
+    // crash diagnostics for pool boxes without a debugger
+    signal(SIGSEGV, [](int sig) {
+      void* frames[64];
+      int n = backtrace(frames, 64);
+      fprintf(stderr, "[etransport] FATAL signal %d, backtrace:\n", sig);
+      backtrace_symbols_fd(frames, n, 2);
+      signal(sig, SIG_DFL);
+      raise(sig);
+    });
+    signal(SIGABRT, [](int sig) {
+      void* frames[64];
+      int n = backtrace(frames, 64);
+      fprintf(stderr, "[etransport] FATAL signal %d, backtrace:\n", sig);
+      backtrace_symbols_fd(frames, n, 2);
+      signal(sig, SIG_DFL);
+      raise(sig);
+    });
+  }
   m.doc() = "C++ data plane for the egrpc server";
+  py::class_<ClientCore>(m, "ClientCore")
+      .def(py::init<const std::string&>())
+      .def("connect", &ClientCore::connect, py::call_guard<py::gil_scoped_release>())
+      .def("connected", &ClientCore::connected)
+      .def("close", &ClientCore::close_fd)
+      .def("call_unary", &ClientCore::call_unary);
   py::class_<ServerCore>(m, "ServerCore")
       .def(py::init<>())
       .def("add_handler", &ServerCore::add_handler)

@@ -83,10 +83,21 @@ class GPUOperator:
 
     @staticmethod
     def _force_symlink(target: str, link: str) -> None:
-        try:
-            os.symlink(target, link)
-        except FileExistsError:
-            if os.path.islink(link) and os.readlink(link) == target:
+        # bounded retry: concurrent binds of the same device-set hash and GC
+        # passes may create/unlink the same link (two pods requesting an
+        # identical ID set share a hash by construction)
+        for _ in range(5):
+            try:
+                os.symlink(target, link)
                 return
-            os.unlink(link)
-            os.symlink(target, link)
+            except FileExistsError:
+                try:
+                    if os.readlink(link) == target:
+                        return
+                except OSError:
+                    pass
+                try:
+                    os.unlink(link)
+                except FileNotFoundError:
+                    pass
+        raise OSError(f"could not materialize symlink {link} -> {target}")

@@ -56,6 +56,10 @@ class GPUSharePlugin:
         self._stop.set()
         self.core_server.stop()
         self.memory_server.stop()
+        # wake any ListAndWatch watcher generators blocked on the refresh
+        # event so their threads exit now rather than at the next interval
+        self.core.trigger_refresh()
+        self.memory.trigger_refresh()
 
     # ---- GC ----
     def gc_once(self) -> int:

@@ -190,6 +190,11 @@ class DevicePluginServer:
         if self._server is not None:
             self._server.stop(grace=0.2)
             self._server = None
+            # unblock streaming generators promptly (their contexts are
+            # cancelled by the server stop; the refresh event wakes them)
+            trigger = getattr(self.plugin, "trigger_refresh", None)
+            if trigger is not None:
+                trigger()
         self._ready.clear()
 
     def start(self) -> None:

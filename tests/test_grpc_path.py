@@ -114,7 +114,7 @@ def test_reregister_after_kubelet_restart(h):
         kubelet2 = FakeKubeletRegistration(h.paths.kubelet_socket)
         kubelet2.start()
         try:
-            assert kubelet2.wait_for_register(1, timeout=15.0)
+            assert kubelet2.wait_for_register(1, timeout=40.0)
             assert len(kubelet2.requests) >= 1
         finally:
             kubelet2.stop()

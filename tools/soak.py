@@ -80,8 +80,12 @@ def main():
             name = f"soak-{wid}-{i}"
             try:
                 if i % 2 == 0:
+                    # each worker owns a disjoint 25-slot range per GPU: real
+                    # kubelet never assigns one device ID to two live pods,
+                    # so concurrent pods must not share ID sets (= hashes)
                     pct = 5 + (i % 4) * 5
-                    start = (wid * 11 + i) % (100 - pct)
+                    base = (wid % 4) * 25
+                    start = base + (i % (25 - 20))
                     ids = [f"{gpu}-{(start + k):02d}" for k in range(pct)]
                     d = Device.new(ids, consts.RESOURCE_GPU_CORE)
                     h.core_locator.assign(d.hash, PodContainer("soak", name, "c"))
@@ -92,7 +96,7 @@ def main():
                     client_core.pre_start({"devicesIDs": ids})
                 else:
                     units = 4 + i % 8
-                    base = (wid * 1000 + i * 17) % 250
+                    base = wid * 300 + (i * 17) % 250  # worker-unique: no cross-worker hash reuse
                     ids = [f"{gpu}-{base + k:06d}" for k in range(units)]
                     d = Device.new(ids, consts.RESOURCE_GPU_MEMORY)
                     h.mem_locator.assign(d.hash, PodContainer("soak", name, "c"))

@@ -353,6 +353,13 @@ class Connection : public std::enable_shared_from_this<Connection> {
         }
         if (stp) dispatch(*stp);
       }
+      // bound per-connection recv-deficit bookkeeping: entries accumulate one
+      // per stream id; safe to drop whenever no stream is open (conn thread
+      // owns this map exclusively)
+      if (stream_recv_deficit_.size() > 4096) {
+        std::lock_guard<std::mutex> lk(streams_mu_);
+        if (streams_.empty()) stream_recv_deficit_.clear();
+      }
     }
   }
 

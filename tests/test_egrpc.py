@@ -25,8 +25,12 @@ def stream_handler(request: bytes, context):
         yield f"chunk-{i}".encode() + request
 
 
-@pytest.fixture
-def eserver(tmp_path):
+@pytest.fixture(params=["native", "python"])
+def eserver(request, tmp_path, monkeypatch):
+    """Differential fixture: every test in this file runs against BOTH the
+    C++ data plane (default) and the pure-Python reference implementation."""
+    if request.param == "python":
+        monkeypatch.setenv("EGPU_PY_TRANSPORT", "1")
     sock = str(tmp_path / "egrpc.sock")
     s = egrpc.Server()
     s.add_service("t.Test", {
@@ -244,7 +248,9 @@ def test_client_reconnects_after_server_restart(tmp_path):
     ch.close()
 
 
-def test_latency_beats_grpcio_floor(eserver, tmp_path):
+def test_latency_beats_grpcio_floor(eserver, request, tmp_path):
+    if "python" in request.node.name:
+        pytest.skip("latency bound targets the native data plane")
     """The entire point: ~100-300µs p50 round trips where grpcio costs ~1ms+.
 
     CI containers are noisy, so the bound is relative: a raw unix-socket

@@ -13,6 +13,8 @@
 
 #include <pybind11/pybind11.h>
 
+#include <cxxabi.h>
+
 #include <algorithm>
 #include <arpa/inet.h>
 #include <atomic>
@@ -313,8 +315,11 @@ class Connection : public std::enable_shared_from_this<Connection> {
   void run() {
     try {
       run_inner();
+    } catch (abi::__forced_unwind&) {
+      // pthread_exit during interpreter finalization: must propagate
+      throw;
     } catch (...) {
-      // defensive: no exception may escape a detached thread
+      // defensive: no other exception may escape a detached thread
     }
     finish();
   }
@@ -737,6 +742,8 @@ class Connection : public std::enable_shared_from_this<Connection> {
       std::thread([self, pack, sid]() {
         try {
           self->run_streaming(pack->fn, pack->ctx, pack->request, sid);
+        } catch (abi::__forced_unwind&) {
+          throw;  // pthread_exit (interpreter finalization)
         } catch (...) {
           self->close_now();
         }
@@ -1008,6 +1015,8 @@ class ServerCore {
       std::thread([conn]() {
         try {
           conn->run();
+        } catch (abi::__forced_unwind&) {
+          throw;  // pthread_exit (interpreter finalization)
         } catch (...) {
           // a stray exception in a detached thread would std::terminate the
           // whole process; drop the connection instead

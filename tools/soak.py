@@ -108,6 +108,10 @@ def main():
                 with lock:
                     counts["prestart"] += 1
                 h.sitter.remove("soak", name)
+                # harness hygiene: each worker cleans its own locator entry
+                # (kubelet's bookkeeping stand-in) so RSS reflects the agent
+                locator = h.core_locator if i % 2 == 0 else h.mem_locator
+                locator.table.pop(d.hash, None)
             except Exception as e:
                 errors.append(f"worker {wid} iter {i}: {e!r}")
                 return

@@ -38,6 +38,14 @@ def main(argv=None) -> int:
         help="disable HSA-shim CU-mask/HBM-quota injection",
     )
     parser.add_argument("--metrics-port", type=int, default=0, help="Prometheus port (0=off)")
+    # path overrides (defaults match the DaemonSet mounts; overridable for
+    # local runs and tests)
+    parser.add_argument("--plugin-dir", default=None, help="kubelet device-plugins dir")
+    parser.add_argument("--podresources-socket", default=None)
+    parser.add_argument("--dev-root", default=None, help="host /dev mount")
+    parser.add_argument("--limits-dir", default=None)
+    parser.add_argument("--state-dir", default=None)
+    parser.add_argument("--shim-host-path", default=None)
     parser.add_argument("-v", "--verbose", action="count", default=0)
     args = parser.parse_args(argv)
 
@@ -48,7 +56,22 @@ def main(argv=None) -> int:
     faulthandler.register(signal.SIGUSR1, all_threads=True)
 
     from ..manager import GPUManager, ManagerOptions
-    from ..plugins.config import PluginOptions
+    from ..plugins.config import AgentPaths, PluginOptions
+
+    paths = AgentPaths()
+    if args.plugin_dir:
+        paths.plugin_dir = args.plugin_dir
+        paths.kubelet_socket = None  # re-derive from plugin_dir
+    if args.podresources_socket:
+        paths.podresources_socket = args.podresources_socket
+    if args.dev_root:
+        paths.dev_root = args.dev_root
+    if args.limits_dir:
+        paths.limits_dir = args.limits_dir
+    if args.state_dir:
+        paths.state_dir = args.state_dir
+    if args.shim_host_path:
+        paths.shim_host_path = args.shim_host_path
 
     opts = ManagerOptions(
         node_name=args.nodeName,
@@ -56,6 +79,7 @@ def main(argv=None) -> int:
         kubeconf=args.kubeconf,
         gpu_plugin_name=args.gpuPluginName,
         backend=args.backend,
+        paths=paths,
         plugin_options=PluginOptions(
             mem_unit_mib=args.mem_unit_mib, isolation=not args.no_isolation
         ),

@@ -1,0 +1,120 @@
+"""End-to-end test of the REAL agent entrypoint as a subprocess:
+`python -m elastic_gpu_agent_amd.cli.agent` with a kubeconfig pointing at the
+stub k8s API server, a fake kubelet registration socket and an in-process
+podresources server — registration, allocation round trip over the wire, and
+clean SIGTERM shutdown."""
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+import pytest
+
+from elastic_gpu_agent_amd import consts
+from elastic_gpu_agent_amd.kube.podresources_server import PodResourcesServer
+from elastic_gpu_agent_amd.types import Device
+
+from helpers import FakeKubeletRegistration, PluginClient
+from test_kube_client import StubK8s
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.timeout(180)
+def test_agent_cli_end_to_end(tmp_path):
+    stub = StubK8s()
+    plugin_dir = tmp_path / "device-plugins"
+    plugin_dir.mkdir()
+    podres_sock = str(tmp_path / "podresources.sock")
+    podres = PodResourcesServer(podres_sock)
+    podres.start()
+    kubelet = FakeKubeletRegistration(str(plugin_dir / "kubelet.sock"))
+    kubelet.start()
+
+    kubeconf = tmp_path / "kubeconfig"
+    kubeconf.write_text(f"""
+apiVersion: v1
+kind: Config
+current-context: ctx
+contexts: [{{name: ctx, context: {{cluster: c, user: u}}}}]
+clusters: [{{name: c, cluster: {{server: "http://127.0.0.1:{stub.port}", insecure-skip-tls-verify: true}}}}]
+users: [{{name: u, user: {{}}}}]
+""")
+
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "elastic_gpu_agent_amd.cli.agent",
+         "--nodeName", "n1",
+         "--dbFile", str(tmp_path / "meta.db"),
+         "--kubeconf", str(kubeconf),
+         "--backend", "fake",
+         "--mem-unit-mib", "1024",
+         "--plugin-dir", str(plugin_dir),
+         "--podresources-socket", podres_sock,
+         "--dev-root", str(tmp_path / "dev"),
+         "--limits-dir", str(tmp_path / "limits"),
+         "--state-dir", str(tmp_path / "state"),
+         "--shim-host-path", str(tmp_path / "libegpu_shim.so")],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        env={**os.environ, "EGPU_FAKE_GPUS": "2"},
+    )
+    try:
+        # both resources register with the fake kubelet
+        assert kubelet.wait_for_register(2, timeout=60), "agent did not register"
+        resources = {r["resource_name"] for r in kubelet.requests}
+        assert resources == {consts.RESOURCE_GPU_CORE, consts.RESOURCE_GPU_MEMORY}
+
+        # full allocate + prestart against the live agent process
+        core_sock = str(plugin_dir / consts.CORE_SOCK_NAME)
+        client = PluginClient(core_sock)
+        try:
+            stream = client.list_and_watch({})
+            first = next(stream)
+            assert len(first["devices"]) == 200  # 2 fake GPUs × 100
+            stream.close()
+
+            ids = [f"1-{i:02d}" for i in range(30)]
+            d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+            pod = stub.add_pod(
+                "default", "cli-pod", node="n1",
+                annotations={
+                    consts.ELASTIC_GPU_ASSUMED_ANNOTATION: "true",
+                    consts.ELASTIC_GPU_CONTAINER_ANNOTATION % "main": "1",
+                })
+            stub.push_event("ADDED", pod)
+            for did in ids:  # >=1.21 podresources shape
+                podres.set_assignment("default", "cli-pod", "main",
+                                      consts.RESOURCE_GPU_CORE, [did])
+            resp = client.allocate({"container_requests": [{"devicesIDs": ids}]})
+            assert resp["container_responses"][0]["envs"]["GPU"] == d.hash
+
+            deadline = time.time() + 30
+            last_err = None
+            while time.time() < deadline:
+                try:
+                    client.pre_start({"devicesIDs": ids})
+                    last_err = None
+                    break
+                except Exception as e:  # sitter may not have synced yet
+                    last_err = e
+                    time.sleep(0.3)
+            assert last_err is None, f"prestart never succeeded: {last_err}"
+            link = tmp_path / "dev" / f"elastic-gpu-{d.hash}-0"
+            assert os.readlink(link) == "/dev/dri/renderD129"
+            limits = json.loads((tmp_path / "limits" / f"{d.hash}.json").read_text())
+            assert limits["cu_count"] == 78  # 30% of 256 → 76.8 → 78 (39 pairs)
+        finally:
+            client.close()
+
+        # clean shutdown on SIGTERM
+        proc.send_signal(signal.SIGTERM)
+        rc = proc.wait(timeout=30)
+        assert rc == 0, (rc, proc.stdout.read()[-2000:])
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+            proc.wait()
+        kubelet.stop()
+        podres.stop()
+        stub.stop()

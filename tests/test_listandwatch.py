@@ -85,3 +85,36 @@ def test_trigger_refresh_resends(h):
     assert got and len(got[0]["devices"]) == 200
     ctx.active = False
     gen.close()
+
+
+def test_mib_unit_snapshot_through_grpcio(tmp_path):
+    """294,912-device ListAndWatch snapshot (1 GPU at 1-MiB units, ~7 MB on
+    the wire) streamed by the native server and decoded by the real gRPC
+    stack — the scale the default memory-unit contract implies."""
+    import grpc
+
+    from helpers import GrpcioPluginClient, Harness
+
+    h = Harness(str(tmp_path), gpus=1, mem_unit_mib=1)
+    try:
+        h.plugin.memory_server.serve()
+        h.plugin.memory_server.wait_ready()
+        ch = grpc.insecure_channel(
+            f"unix://{h.plugin.memory_server.socket_path}",
+            options=[("grpc.max_receive_message_length", 64 << 20)],
+        )
+        from elastic_gpu_agent_amd.protos import deviceplugin as dp
+
+        stream = ch.unary_stream(
+            dp.METHOD_LIST_AND_WATCH,
+            request_serializer=dp.Empty.encode,
+            response_deserializer=dp.ListAndWatchResponse.decode,
+        )({})
+        first = next(stream)
+        assert len(first["devices"]) == 288 * 1024  # 294,912
+        assert first["devices"][0]["ID"] == "0-000000"
+        assert first["devices"][-1]["ID"] == f"0-{288 * 1024 - 1:06d}"
+        stream.cancel()
+        ch.close()
+    finally:
+        h.close()

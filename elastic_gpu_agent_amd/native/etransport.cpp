@@ -1341,6 +1341,24 @@ class ClientCore {
 
 }  // namespace
 
+// test hook: the C++ HPACK decoder exposed for the differential fuzz suite
+class HpackTester {
+ public:
+  py::list decode(py::bytes data) {
+    std::string raw = data;
+    std::vector<std::pair<std::string, std::string>> out;
+    if (!dec_.decode((const uint8_t*)raw.data(), raw.size(), &out))
+      throw std::runtime_error("hpack decode failed");
+    py::list result;
+    for (auto& h : out)
+      result.append(py::make_tuple(py::bytes(h.first), py::bytes(h.second)));
+    return result;
+  }
+
+ private:
+  HpackDecoder dec_;
+};
+
 PYBIND11_MODULE(_etransport, m) {
   trie_init();
   if (const char* dbg = getenv("EGPU_ETRANSPORT_DEBUG"); dbg && dbg[0] == '1') {
@@ -1363,6 +1381,9 @@ PYBIND11_MODULE(_etransport, m) {
     });
   }
   m.doc() = "C++ data plane for the egrpc server";
+  py::class_<HpackTester>(m, "HpackTester")
+      .def(py::init<>())
+      .def("decode", &HpackTester::decode);
   py::class_<ClientCore>(m, "ClientCore")
       .def(py::init<const std::string&>())
       .def("connect", &ClientCore::connect, py::call_guard<py::gil_scoped_release>())

@@ -132,3 +132,118 @@ def test_huffman_table_against_nghttp2_inflater():
               b"grpc-go/1.27.0", b"trailers", b"0", b"elasticgpu.io/gpu-core"):
         n, v = inflate_literal(b"p", s)
         assert v == s
+
+
+@pytest.mark.skipif(not os.path.exists(NGHTTP2), reason="libnghttp2 not present")
+def test_differential_fuzz_against_nghttp2_deflater():
+    """Differential fuzz: random header lists encoded by nghttp2's HPACK
+    deflater (which uses indexed entries, dynamic-table references, Huffman
+    and table-size updates like real kubelet/grpc encoders) must decode
+    identically through our connection-scoped decoder — including dynamic
+    table state carried ACROSS blocks on one connection."""
+    lib = ctypes.CDLL(NGHTTP2)
+    lib.nghttp2_hd_deflate_new.argtypes = [ctypes.POINTER(ctypes.c_void_p), ctypes.c_size_t]
+    lib.nghttp2_hd_deflate_hd.restype = ctypes.c_ssize_t
+    lib.nghttp2_hd_deflate_hd.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8), ctypes.c_size_t,
+        ctypes.POINTER(_NV), ctypes.c_size_t,
+    ]
+
+    rng = random.Random(20260913)
+    names = [b":path", b":method", b":authority", b"content-type", b"te",
+             b"grpc-timeout", b"user-agent", b"x-custom-header", b"x-trace-id",
+             b"grpc-encoding", b"authorization"]
+    values = [b"POST", b"/v1beta1.DevicePlugin/Allocate", b"application/grpc",
+              b"trailers", b"10S", b"grpc-go/1.27.1 (linux; amd64)", b"",
+              b"elasticgpu.io/gpu-core", b"0", b"identity"]
+
+    def random_headers():
+        out = []
+        for _ in range(rng.randrange(1, 10)):
+            if rng.random() < 0.7:
+                n = rng.choice(names)
+            else:
+                n = bytes(rng.choice(b"abcdefghijklmnopqrstuvwxyz-")
+                          for _ in range(rng.randrange(1, 20)))
+            if rng.random() < 0.6:
+                v = rng.choice(values)
+            else:
+                v = bytes(rng.randrange(0x20, 0x7F)
+                          for _ in range(rng.randrange(0, 40)))
+            out.append((n, v))
+        return out
+
+    for conn_round in range(30):  # 30 fresh "connections"
+        deflater = ctypes.c_void_p()
+        assert lib.nghttp2_hd_deflate_new(ctypes.byref(deflater), 4096) == 0
+        dec = hpack.Decoder()
+        for block_round in range(10):  # 10 header blocks per connection
+            headers = random_headers()
+            nva = (_NV * len(headers))()
+            keepalive = []
+            for i, (n, v) in enumerate(headers):
+                nb, vb = ctypes.create_string_buffer(n, len(n)), \
+                    ctypes.create_string_buffer(v, len(v))
+                keepalive += [nb, vb]
+                nva[i].name = ctypes.cast(nb, ctypes.c_void_p)
+                nva[i].value = ctypes.cast(vb, ctypes.c_void_p)
+                nva[i].namelen = len(n)
+                nva[i].valuelen = len(v)
+                nva[i].flags = 0
+            buf = (ctypes.c_uint8 * 65536)()
+            rv = lib.nghttp2_hd_deflate_hd(deflater, buf, 65536, nva, len(headers))
+            assert rv > 0, f"deflate failed rv={rv}"
+            block = bytes(buf[:rv])
+            decoded = dec.decode(block)
+            assert decoded == headers, (
+                f"conn {conn_round} block {block_round}: {decoded} != {headers}"
+            )
+        lib.nghttp2_hd_deflate_del(deflater)
+
+
+@pytest.mark.skipif(not os.path.exists(NGHTTP2), reason="libnghttp2 not present")
+def test_cpp_decoder_differential_fuzz():
+    """The C++ transport core's HPACK decoder (native/etransport.cpp) under
+    the same nghttp2-deflater fuzz as the Python decoder."""
+    try:
+        from elastic_gpu_agent_amd import _etransport
+    except ImportError:
+        pytest.skip("_etransport not built")
+    lib = ctypes.CDLL(NGHTTP2)
+    lib.nghttp2_hd_deflate_new.argtypes = [ctypes.POINTER(ctypes.c_void_p), ctypes.c_size_t]
+    lib.nghttp2_hd_deflate_hd.restype = ctypes.c_ssize_t
+    lib.nghttp2_hd_deflate_hd.argtypes = [
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint8), ctypes.c_size_t,
+        ctypes.POINTER(_NV), ctypes.c_size_t,
+    ]
+    rng = random.Random(777)
+    for conn_round in range(20):
+        deflater = ctypes.c_void_p()
+        assert lib.nghttp2_hd_deflate_new(ctypes.byref(deflater), 4096) == 0
+        dec = _etransport.HpackTester()
+        for block_round in range(10):
+            headers = [
+                (bytes(rng.choice(b"abcdefgh-:") for _ in range(rng.randrange(1, 12))),
+                 bytes(rng.randrange(0x20, 0x7F) for _ in range(rng.randrange(0, 30))))
+                for _ in range(rng.randrange(1, 8))
+            ]
+            # mix in realistic repeats that exercise dynamic-table indexing
+            headers += [(b":path", b"/v1beta1.DevicePlugin/Allocate"),
+                        (b"content-type", b"application/grpc")]
+            nva = (_NV * len(headers))()
+            keepalive = []
+            for i, (n, v) in enumerate(headers):
+                nb = ctypes.create_string_buffer(n, len(n))
+                vb = ctypes.create_string_buffer(v, len(v))
+                keepalive += [nb, vb]
+                nva[i].name = ctypes.cast(nb, ctypes.c_void_p)
+                nva[i].value = ctypes.cast(vb, ctypes.c_void_p)
+                nva[i].namelen = len(n)
+                nva[i].valuelen = len(v)
+                nva[i].flags = 0
+            buf = (ctypes.c_uint8 * 65536)()
+            rv = lib.nghttp2_hd_deflate_hd(deflater, buf, 65536, nva, len(headers))
+            assert rv > 0
+            decoded = dec.decode(bytes(buf[:rv]))
+            assert decoded == headers, f"conn {conn_round} block {block_round}"
+        lib.nghttp2_hd_deflate_del(deflater)

@@ -43,21 +43,23 @@ class GPUSharePluginBase:
     def fake_device_ids_for_gpu(self, gpu) -> List[str]:
         raise NotImplementedError
 
-    def list_devices(self) -> List[dict]:
+    def list_devices(self, healthy: bool = True) -> List[dict]:
+        health = consts.HEALTHY if healthy else consts.UNHEALTHY
         out = []
         for gpu in self.cfg.operator.devices():
             topo = {"nodes": [{"ID": gpu.numa_node}]}
             for did in self.fake_device_ids_for_gpu(gpu):
-                out.append({"ID": did, "health": consts.HEALTHY, "topology": topo})
+                out.append({"ID": did, "health": health, "topology": topo})
         return out
 
-    def device_groups(self) -> List[tuple]:
+    def device_groups(self, healthy: bool = True) -> List[tuple]:
         """[(ids, encoded Device suffix)] per GPU — the fast-encode shape."""
         from ..protos import fastpath
 
+        health = consts.HEALTHY if healthy else consts.UNHEALTHY
         groups = []
         for gpu in self.cfg.operator.devices():
-            suffix = fastpath.device_suffix(consts.HEALTHY, gpu.numa_node)
+            suffix = fastpath.device_suffix(health, gpu.numa_node)
             groups.append((self.fake_device_ids_for_gpu(gpu), suffix))
         return groups
 
@@ -65,15 +67,20 @@ class GPUSharePluginBase:
     def get_device_plugin_options(self, request, context) -> dict:
         return {"pre_start_required": True, "get_preferred_allocation_available": True}
 
+    UNHEALTHY_AFTER_FAILURES = 3
+
     def _watch_snapshots(self, context, snap_fn):
         """Initial snapshot + re-advertisement when enumeration changes.
 
         Unlike the reference (single static send, devices never re-checked —
-        SURVEY §3.2), the backend is re-enumerated periodically so a GPU
-        falling off the bus transitions its fake devices out of the list."""
-        current = snap_fn()
+        SURVEY §3.2), the backend is re-enumerated periodically: a GPU
+        falling off the bus shrinks the advertised list, and when enumeration
+        itself keeps failing (driver wedged, amdsmi gone) every device is
+        re-advertised Unhealthy so kubelet stops placing pods here."""
+        current = snap_fn(True)
         yield current
         interval = self.cfg.options.health_refresh_seconds
+        failures = 0
         while context is None or context.is_active():
             triggered = self._refresh.wait(timeout=interval)
             self._refresh.clear()
@@ -81,18 +88,26 @@ class GPUSharePluginBase:
                 return
             try:
                 self.cfg.operator.devices(refresh=True)
+                recovered = failures >= self.UNHEALTHY_AFTER_FAILURES
+                failures = 0
             except Exception as e:
-                log.error("device re-enumeration failed: %s", e)
+                failures += 1
+                log.error("device re-enumeration failed (%d): %s", failures, e)
+                if failures == self.UNHEALTHY_AFTER_FAILURES:
+                    current = snap_fn(False)
+                    yield current
                 continue
-            fresh = snap_fn()
-            if fresh != current:
+            fresh = snap_fn(True)
+            if fresh != current or recovered:
                 current = fresh
                 yield current
             elif triggered:
                 yield current
 
     def list_and_watch(self, context):
-        yield from self._watch_snapshots(context, lambda: {"devices": self.list_devices()})
+        yield from self._watch_snapshots(
+            context, lambda healthy: {"devices": self.list_devices(healthy)}
+        )
 
     def list_and_watch_encoded(self, context):
         """Server path: pre-encoded ListAndWatchResponse bytes (cached until
@@ -101,7 +116,8 @@ class GPUSharePluginBase:
         from ..protos import fastpath
 
         yield from self._watch_snapshots(
-            context, lambda: fastpath.encode_list_and_watch(self.device_groups())
+            context,
+            lambda healthy: fastpath.encode_list_and_watch(self.device_groups(healthy)),
         )
 
     def trigger_refresh(self) -> None:

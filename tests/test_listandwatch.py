@@ -118,3 +118,46 @@ def test_mib_unit_snapshot_through_grpcio(tmp_path):
         ch.close()
     finally:
         h.close()
+
+
+def test_unhealthy_after_repeated_enumeration_failure(h):
+    """3 consecutive enumeration failures → everything re-advertised
+    Unhealthy; recovery → Healthy again."""
+    ctx = _Ctx()
+    gen = h.plugin.core.list_and_watch(ctx)
+    first = next(gen)
+    assert all(d["health"] == "Healthy" for d in first["devices"])
+
+    real_devices = h.operator.backend.devices
+
+    def boom():
+        raise RuntimeError("amdsmi wedged")
+
+    got = []
+
+    def consume(n):
+        for resp in gen:
+            got.append(resp)
+            if len(got) >= n:
+                break
+
+    h.operator.backend.devices = boom
+    t = threading.Thread(target=lambda: consume(1), daemon=True)
+    t.start()
+    deadline = time.time() + 10
+    while not got and time.time() < deadline:
+        time.sleep(0.05)
+    assert got, "no Unhealthy re-advertisement"
+    assert all(d["health"] == "Unhealthy" for d in got[0]["devices"])
+
+    # recovery
+    h.operator.backend.devices = real_devices
+    t2 = threading.Thread(target=lambda: consume(2), daemon=True)
+    t2.start()
+    deadline = time.time() + 10
+    while len(got) < 2 and time.time() < deadline:
+        time.sleep(0.05)
+    assert len(got) >= 2, "no recovery re-advertisement"
+    assert all(d["health"] == "Healthy" for d in got[1]["devices"])
+    ctx.active = False
+    gen.close()

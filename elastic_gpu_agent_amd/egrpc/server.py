@@ -548,8 +548,11 @@ class Server:
             threading.Thread(target=conn.run, name="egrpc-conn", daemon=True).start()
 
     def stop(self, grace: float = 0.0) -> None:
-        if self._native is not None:
-            self._native.stop()
+        native = self._native
+        if native is not None:
+            # native stop is idempotent/thread-safe; keep the reference until
+            # it returns so concurrent stop() callers never race the teardown
+            native.stop()
             self._native = None
             return
         self._stopped.set()

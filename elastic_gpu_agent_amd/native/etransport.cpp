@@ -975,6 +975,11 @@ class ServerCore {
   }
 
   void stop() {
+    // idempotent + thread-safe: shutdown paths (signal handler, watch loop,
+    // test teardown) may race
+    std::lock_guard<std::mutex> stop_lk(stop_mu_);
+    if (stopped_) return;
+    stopped_ = true;
     core_->stopping.store(true);
     if (listen_fd_ >= 0) {
       ::shutdown(listen_fd_, SHUT_RDWR);
@@ -1032,6 +1037,8 @@ class ServerCore {
   std::thread accept_thread_;
   std::mutex conns_mu_;
   std::vector<std::shared_ptr<Connection>> conns_;
+  std::mutex stop_mu_;
+  bool stopped_ = false;
 };
 
 

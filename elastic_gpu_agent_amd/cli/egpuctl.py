@@ -5,8 +5,11 @@ Subcommands:
   pods                    list persisted allocations
   masks                   list live CU-mask assignments
   occupancy               per-pod live occupancy (amdsmi join)
-  gc                      run one GC reconciliation pass (needs cluster access)
+  partition IDX [MODE]    get/set SPX|DPX|QPX|CPX compute partition
   migrate --from BOLT_DB  import a reference agent's BoltDB state
+
+(GC runs inside the agent daemon — event-driven + 60 s timer; there is no
+offline GC because reclamation requires the live pod view.)
 """
 from __future__ import annotations
 

@@ -43,12 +43,10 @@ class PodSitter(Sitter):
         client: K8sClient,
         node_name: str,
         delete_hook: Optional[Callable[[Pod], None]] = None,
-        relist_interval: float = 30.0,
     ):
         self._client = client
         self._node = node_name
         self._hook = delete_hook
-        self._relist = relist_interval
         self._cache: Dict[str, Pod] = {}
         self._lock = threading.Lock()
         self._synced = threading.Event()

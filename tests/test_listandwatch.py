@@ -93,7 +93,7 @@ def test_mib_unit_snapshot_through_grpcio(tmp_path):
     stack — the scale the default memory-unit contract implies."""
     import grpc
 
-    from helpers import GrpcioPluginClient, Harness
+    from helpers import Harness
 
     h = Harness(str(tmp_path), gpus=1, mem_unit_mib=1)
     try:

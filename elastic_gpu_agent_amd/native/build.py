@@ -79,7 +79,7 @@ def build_etransport(force=False):
     if not force and _newer(out, src, tables):
         return out
     _run(
-        ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src]
+        ["g++", "-O2", "-g", "-std=c++17", "-shared", "-fPIC", src]
         + pybind_includes()
         + ["-pthread", "-o", out]
     )

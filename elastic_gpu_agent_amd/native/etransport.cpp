@@ -288,6 +288,7 @@ struct CoreShared {
   py::object context_factory;   // callable() -> ctx
   py::object error_introspect;  // callable(exc) -> (code:int, message:str)
   std::atomic<bool> stopping{false};
+  std::atomic<int> live_threads{0};  // conn + streaming threads in flight
 
   ~CoreShared() {
     // the last shared_ptr may be dropped from a connection thread that does
@@ -739,10 +740,13 @@ class Connection : public std::enable_shared_from_this<Connection> {
       // the thread must keep the Connection alive: a raw `this` would dangle
       // once the client disconnects and the server prunes the connection
       std::shared_ptr<Connection> self = shared_from_this();
-      std::thread([self, pack, sid]() {
+      core_->live_threads.fetch_add(1);
+      auto core = core_;
+      std::thread([self, pack, sid, core]() {
         try {
           self->run_streaming(pack->fn, pack->ctx, pack->request, sid);
         } catch (abi::__forced_unwind&) {
+          core->live_threads.fetch_sub(1);
           throw;  // pthread_exit (interpreter finalization)
         } catch (...) {
           self->close_now();
@@ -751,6 +755,7 @@ class Connection : public std::enable_shared_from_this<Connection> {
           py::gil_scoped_acquire gil;
           delete pack;
         }
+        core->live_threads.fetch_sub(1);
       }).detach();
       return;
     }
@@ -992,6 +997,15 @@ class ServerCore {
     }
     if (accept_thread_.joinable()) accept_thread_.join();
     ::unlink(path_.c_str());
+    // Drain detached connection/streaming threads (they exit promptly once
+    // their sockets are shut down). Without this, interpreter finalization
+    // can force-unwind a thread that holds a pybind GIL scope, whose
+    // destructor then aborts the process. stop() is called WITHOUT the GIL
+    // (call_guard), so in-flight Python handlers can finish.
+    for (int i = 0; i < 500 && core_->live_threads.load() > 0; ++i) {
+      struct timespec ts {0, 10 * 1000 * 1000};  // 10 ms
+      nanosleep(&ts, nullptr);
+    }
   }
 
   ~ServerCore() {
@@ -1017,16 +1031,20 @@ class ServerCore {
               conns_.end());
         }
       }
-      std::thread([conn]() {
+      core_->live_threads.fetch_add(1);
+      auto core = core_;
+      std::thread([conn, core]() {
         try {
           conn->run();
         } catch (abi::__forced_unwind&) {
+          core->live_threads.fetch_sub(1);
           throw;  // pthread_exit (interpreter finalization)
         } catch (...) {
           // a stray exception in a detached thread would std::terminate the
           // whole process; drop the connection instead
           conn->close_now();
         }
+        core->live_threads.fetch_sub(1);
       }).detach();
     }
   }

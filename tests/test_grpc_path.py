@@ -107,10 +107,14 @@ def test_reregister_after_kubelet_restart(h):
         assert kubelet.wait_for_register(1)
         n0 = len(kubelet.requests)
 
-        # kubelet restart: old socket replaced by a new one (new inode)
+        # kubelet restart: old socket replaced by a new one (new inode).
+        # A real kubelet restart takes seconds; recreate after >1 watcher
+        # tick so the (inode, ctime) identity check can't alias a same-tick
+        # recreation with a recycled inode (tmpfs does recycle them).
         kubelet.stop()
         if os.path.exists(h.paths.kubelet_socket):  # grpc may remove it on stop
             os.unlink(h.paths.kubelet_socket)
+        time.sleep(1.5)
         kubelet2 = FakeKubeletRegistration(h.paths.kubelet_socket)
         kubelet2.start()
         try:

@@ -154,9 +154,14 @@ class GPUSharePluginBase:
         if not (self.cfg.options.isolation and self.cfg.limits and paths.shim_host_path):
             return resp
         # NOTE: the limits file itself is written at PreStart (before the
-        # container is created); Allocate only declares the mount. Keeps the
-        # Allocate hot path free of disk I/O (p50 latency is the headline).
-        limits_host = self.cfg.limits.host_path(device.hash)
+        # container is created); Allocate only declares the mount — with the
+        # HOST-view path (kubelet resolves host_path on the host, not through
+        # the agent's /host mount). Keeps the Allocate hot path free of disk
+        # I/O (p50 latency is the headline).
+        import os as _os
+
+        limits_host = paths.limits_host_view(
+            _os.path.basename(self.cfg.limits.host_path(device.hash)))
         resp["mounts"] = [
             {
                 "container_path": paths.shim_container_path,

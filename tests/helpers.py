@@ -50,6 +50,7 @@ class Harness:
             plugin_dir=plugin_dir,
             kubelet_socket=os.path.join(plugin_dir, "kubelet.sock"),
             limits_dir=limits_dir,
+            limits_dir_host=None,  # tests: agent view == host view
             shim_host_path=os.path.join(self.tmp, "libegpu_shim.so"),
         )
         cfg = GPUPluginConfig(

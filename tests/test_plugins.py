@@ -223,3 +223,37 @@ def test_cumask_disjoint_across_pods(h):
         masks.append(parse_mask_hex(h.plugin.cfg.limits.read(d.hash)["cu_mask"]))
     a, b = masks
     assert all((wa & wb) == 0 for wa, wb in zip(a, b))
+
+
+def test_allocate_advertises_host_view_paths(tmp_path):
+    """Production defaults: the agent WRITES through /host/... but must
+    ADVERTISE true host paths in Allocate mounts (kubelet resolves host_path
+    on the host). Regression for the /host-prefix leak."""
+    from elastic_gpu_agent_amd.isolation import LimitsWriter
+    from elastic_gpu_agent_amd.plugins.config import AgentPaths, GPUPluginConfig, PluginOptions
+    from elastic_gpu_agent_amd.plugins.gpushare import GPUShareCorePlugin
+    from elastic_gpu_agent_amd.operator import GPUOperator
+    from elastic_gpu_agent_amd.operator.fake import FakeBackend
+    from elastic_gpu_agent_amd.storage import Storage
+
+    paths = AgentPaths()  # production defaults, except limits written to tmp
+    paths.limits_dir = str(tmp_path / "limits")
+    cfg = GPUPluginConfig(
+        operator=GPUOperator(FakeBackend(count=1), dev_root=str(tmp_path / "dev")),
+        storage=Storage(str(tmp_path / "meta.db")),
+        sitter=None, core_locator=None, memory_locator=None,
+        paths=paths, options=PluginOptions(),
+        limits=LimitsWriter(paths.limits_dir),
+    )
+    plugin = GPUShareCorePlugin(cfg)
+    ids = [f"0-{i:02d}" for i in range(25)]
+    resp = plugin.allocate({"container_requests": [{"devicesIDs": ids}]}, None)
+    cr = resp["container_responses"][0]
+    mounts = {m["container_path"]: m["host_path"] for m in cr["mounts"]}
+    assert mounts["/opt/egpu/libegpu_shim.so"] == "/opt/egpu/libegpu_shim.so"
+    limits_host = mounts["/etc/egpu/limits-core.json"]
+    assert limits_host.startswith("/var/lib/egpu/limits/")
+    assert not limits_host.startswith("/host/")
+    for spec in cr["devices"]:
+        assert not spec["host_path"].startswith("/host/")
+    cfg.storage.close()

@@ -271,6 +271,47 @@ def test_cu_mask_applies_to_torch_kernels(gpus):
     assert ratio > 1.8, f"25% mask gave only {ratio:.2f}x on torch matmul"
 
 
+def test_limits_file_path_enforced(tmp_path, gpus):
+    """The PRODUCTION config channel: the shim reads the limits JSON the
+    agent mounts at /etc/egpu (here via EGPU_LIMITS_DIR), not env overrides —
+    CU mask + HBM quota + QoS priority all flow from the file."""
+    from elastic_gpu_agent_amd.isolation import LimitsWriter
+    from elastic_gpu_agent_amd.isolation.cumask import mask_for_percent, mask_hex
+
+    writer = LimitsWriter(str(tmp_path))
+    words, n_cus = mask_for_percent(25, gpus[0].cu_count, gpus[0].xcd_count)
+    writer.finalize(
+        "demo1234", [gpus[0].index], gpus, cu_mask=mask_hex(words), cu_count=n_cus,
+        mem_limit_bytes=2 * 1024**3, priority="high",
+    )
+    # shim scans limits*.json: give it the in-container naming
+    os.rename(writer.host_path("demo1234"), str(tmp_path / "limits-core.json"))
+
+    env = dict(os.environ)
+    env.update({
+        "HSA_TOOLS_LIB": SHIM,
+        "EGPU_LIMITS_DIR": str(tmp_path),
+        "EGPU_SHIM_VERBOSE": "1",
+        "EGPU_SHIM_DEBUG": "1",
+    })
+    code = (
+        "from elastic_gpu_agent_amd.isolation import probes; import json; "
+        "seen = probes.census(0, blocks=2048, spin=150000); "
+        "over = probes.malloc_bytes(0, 4 * 1024**3); "
+        "ok = probes.malloc_bytes(0, 1024**3); "
+        "print(json.dumps([len(seen), over, ok]))"
+    )
+    out = subprocess.run([sys.executable, "-c", code], env=env, cwd=REPO,
+                         capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-3000:]
+    seen, over, ok = json.loads(out.stdout.strip().splitlines()[-1])
+    assert seen <= n_cus, f"file-config mask not enforced: {seen} > {n_cus}"
+    assert over != 0, "file-config quota not enforced"
+    assert ok == 0, "in-quota alloc failed"
+    # priority parsed and applied to queues (shim logs it)
+    assert "priority -> 2" in out.stderr, out.stderr[-1500:]
+
+
 def test_hook_real_injection(tmp_path, gpus):
     """Non-dry-run hook: mknod into a real separate mount namespace.
 

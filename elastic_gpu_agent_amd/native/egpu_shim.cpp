@@ -196,14 +196,30 @@ AmdExtTable g_amdext{};    // saved original amd-ext entry points
 std::atomic<uint64_t> g_vram_used{0};
 std::atomic<uint64_t> g_denied{0};
 std::atomic<uint64_t> g_queues_masked{0};
-std::mutex g_alloc_mu;
-std::unordered_map<void*, uint64_t> g_alloc_sizes;
+// Deliberately leaked (never destroyed): libamdhip64's atexit teardown
+// frees memory through our wrappers AFTER this DSO's static destructors
+// would have run — destroyed mutexes/maps here were the cause of a
+// post-deny teardown segfault (symbolized to refund()/pool_free_wrap).
+std::mutex& alloc_mu() {
+  static std::mutex* m = new std::mutex();
+  return *m;
+}
+std::unordered_map<void*, uint64_t>& alloc_sizes() {
+  static auto* m = new std::unordered_map<void*, uint64_t>();
+  return *m;
+}
 
 // VRAM pool/region classification (lazily built: agents exist only after
 // hsa_init completes, long before the first allocation).
 std::once_flag g_pools_once;
-std::unordered_set<uint64_t> g_vram_pools;    // hsa_amd_memory_pool_t.handle
-std::unordered_set<uint64_t> g_vram_regions;  // hsa_region_t.handle
+std::unordered_set<uint64_t>& vram_pools() {
+  static auto* s = new std::unordered_set<uint64_t>();
+  return *s;
+}
+std::unordered_set<uint64_t>& vram_regions() {
+  static auto* s = new std::unordered_set<uint64_t>();
+  return *s;
+}
 
 void build_pool_sets() {
   auto agent_cb = [](hsa_agent_t agent, void*) -> hsa_status_t {
@@ -216,7 +232,7 @@ void build_pool_sets() {
       if (g_amdext.hsa_amd_memory_pool_get_info_fn(pool, HSA_AMD_MEMORY_POOL_INFO_SEGMENT, &seg) ==
               HSA_STATUS_SUCCESS &&
           seg == HSA_AMD_SEGMENT_GLOBAL) {
-        g_vram_pools.insert(pool.handle);
+        vram_pools().insert(pool.handle);
       }
       return HSA_STATUS_SUCCESS;
     };
@@ -226,7 +242,7 @@ void build_pool_sets() {
       if (g_core.hsa_region_get_info_fn(region, HSA_REGION_INFO_SEGMENT, &seg) ==
               HSA_STATUS_SUCCESS &&
           seg == HSA_REGION_SEGMENT_GLOBAL) {
-        g_vram_regions.insert(region.handle);
+        vram_regions().insert(region.handle);
       }
       return HSA_STATUS_SUCCESS;
     };
@@ -234,33 +250,34 @@ void build_pool_sets() {
     return HSA_STATUS_SUCCESS;
   };
   g_core.hsa_iterate_agents_fn(agent_cb, nullptr);
-  logf("classified %zu VRAM pools, %zu VRAM regions", g_vram_pools.size(),
-       g_vram_regions.size());
+  logf("classified %zu VRAM pools, %zu VRAM regions", vram_pools().size(),
+       vram_regions().size());
 }
 
 bool is_vram_pool(hsa_amd_memory_pool_t pool) {
   std::call_once(g_pools_once, build_pool_sets);
-  return g_vram_pools.count(pool.handle) != 0;
+  return vram_pools().count(pool.handle) != 0;
 }
 
 bool is_vram_region(hsa_region_t region) {
   std::call_once(g_pools_once, build_pool_sets);
-  return g_vram_regions.count(region.handle) != 0;
+  return vram_regions().count(region.handle) != 0;
 }
 
 void record_ptr(void* ptr, uint64_t size) {
   if (g_cfg.mem_limit == 0) return;
-  std::lock_guard<std::mutex> lk(g_alloc_mu);
-  g_alloc_sizes[ptr] = size;
+  std::lock_guard<std::mutex> lk(alloc_mu());
+  alloc_sizes()[ptr] = size;
 }
 
 void refund(void* ptr) {
   if (g_cfg.mem_limit == 0 || ptr == nullptr) return;
-  std::lock_guard<std::mutex> lk(g_alloc_mu);
-  auto it = g_alloc_sizes.find(ptr);
-  if (it != g_alloc_sizes.end()) {
+  std::lock_guard<std::mutex> lk(alloc_mu());
+  auto& sizes = alloc_sizes();
+  auto it = sizes.find(ptr);
+  if (it != sizes.end()) {
     g_vram_used.fetch_sub(it->second);
-    g_alloc_sizes.erase(it);
+    sizes.erase(it);
   }
 }
 

@@ -11,7 +11,6 @@ from __future__ import annotations
 import logging
 import queue
 import threading
-import time
 from typing import List
 
 from .. import consts

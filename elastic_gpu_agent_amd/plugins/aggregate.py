@@ -25,6 +25,21 @@ log = logging.getLogger(__name__)
 GC_PERIOD_SECONDS = 60.0
 
 
+def _malloc_trim() -> None:
+    """Return glibc's freed-but-retained heap pages to the kernel.
+
+    Under sustained RPC churn across threads, glibc keeps freed chunks
+    resident (bins/fragmentation; MALLOC_ARENA_MAX-insensitive) — measured
+    ~0.3-1 KB per pod cycle of RSS creep, fully reclaimed by malloc_trim
+    (soak plateaus flat with this in the GC cadence; see docs/TESTING.md)."""
+    try:
+        import ctypes
+
+        ctypes.CDLL("libc.so.6").malloc_trim(0)
+    except Exception:  # non-glibc platforms: harmless to skip
+        pass
+
+
 class GPUSharePlugin:
     def __init__(self, config: GPUPluginConfig):
         self.cfg = config
@@ -121,6 +136,7 @@ class GPUSharePlugin:
                 self.gc_once()
             except Exception as e:
                 log.error("GC pass failed: %s", e)
+            _malloc_trim()
 
     # ---- Restore ----
     def restore(self) -> int:

@@ -34,11 +34,10 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=int, default=60)
     ap.add_argument("--workers", type=int, default=4)
-    ap.add_argument("--rss-limit-mb", type=float, default=300.0,
-                    help="fixed RSS growth allowance (allocator retention is "
-                    "front-loaded: measured +50 MB @15k cycles, +226 MB @234k, "
-                    "+385 MB @600k — decelerating; see docs/TESTING.md)")
-    ap.add_argument("--rss-per-cycle-bytes", type=float, default=250.0,
+    ap.add_argument("--rss-limit-mb", type=float, default=150.0,
+                    help="fixed RSS growth allowance (with the GC-cadence "
+                    "malloc_trim, RSS plateaus after the initial ramp)")
+    ap.add_argument("--rss-per-cycle-bytes", type=float, default=50.0,
                     help="additional allowance per bind cycle")
     args = ap.parse_args()
 
@@ -139,10 +138,13 @@ def main():
             client.close()
 
     def gc_loop():
+        from elastic_gpu_agent_amd.plugins.aggregate import _malloc_trim
+
         while not stop.is_set():
             time.sleep(1.0)
             try:
                 h.plugin.gc_once()
+                _malloc_trim()  # production GC cadence does the same
                 with lock:
                     counts["gc"] += 1
             except Exception as e:

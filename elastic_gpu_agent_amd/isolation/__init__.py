@@ -40,19 +40,30 @@ class CUMaskAllocator:
         self._storage = storage
         self._devices = {d.index: d for d in devices}
         self._lock = threading.Lock()
+        # In-memory occupancy cache, rebuilt from the aux table at startup
+        # (restart safety) and maintained on allocate/release: per-allocation
+        # work must not scan every live record (O(n²) under churn).
+        self._used: Dict[int, set] = {}
+        self._by_hash: Dict[str, tuple] = {}  # hash -> (gpu_index, set_of_cus)
+        for key, val in storage.aux_items(AUX_MASK_PREFIX):
+            rec = json.loads(val)
+            cus = self._mask_cus(rec["cu_mask"])
+            gpu = rec.get("gpu_index")
+            self._by_hash[key[len(AUX_MASK_PREFIX):]] = (gpu, cus)
+            self._used.setdefault(gpu, set()).update(cus)
+
+    @staticmethod
+    def _mask_cus(hexmask: str) -> set:
+        cus = set()
+        for w_i, w in enumerate(parse_mask_hex(hexmask)):
+            for b in range(32):
+                if w >> b & 1:
+                    cus.add(w_i * 32 + b)
+        return cus
 
     # ---- occupancy ----
     def _live_cus(self, gpu_index: int) -> set:
-        used = set()
-        for key, val in self._storage.aux_items(AUX_MASK_PREFIX):
-            rec = json.loads(val)
-            if rec.get("gpu_index") == gpu_index:
-                words = parse_mask_hex(rec["cu_mask"])
-                for w_i, w in enumerate(words):
-                    for b in range(32):
-                        if w >> b & 1:
-                            used.add(w_i * 32 + b)
-        return used
+        return self._used.setdefault(gpu_index, set())
 
     def allocate(self, alloc_hash: str, gpu_index: int, percent: int) -> Tuple[str, int]:
         """Pick a CU set of ``percent``% of the GPU, disjoint from live
@@ -124,10 +135,39 @@ class CUMaskAllocator:
                     }
                 ),
             )
+            cu_set = set(cus)
+            old = self._by_hash.pop(alloc_hash, None)
+            if old is not None:  # re-allocation of the same hash
+                self._used.setdefault(old[0], set()).difference_update(old[1])
+            self._by_hash[alloc_hash] = (gpu_index, cu_set)
+            self._used.setdefault(gpu_index, set()).update(cu_set)
         return hexmask, n_eff
 
+    def _release_cached(self, alloc_hash: str) -> None:
+        entry = self._by_hash.pop(alloc_hash, None)
+        if entry is None:
+            return
+        gpu_index, cus = entry
+        used = self._used.setdefault(gpu_index, set())
+        # only remove CUs not still claimed by another live mask
+        still = set()
+        for g, c in self._by_hash.values():
+            if g == gpu_index:
+                still |= c
+        used.difference_update(cus - still)
+
     def release(self, alloc_hash: str) -> None:
+        with self._lock:
+            self._release_cached(alloc_hash)
         self._storage.aux_delete(AUX_MASK_PREFIX + alloc_hash)
+
+    def release_many(self, hashes) -> None:
+        """Batch form for GC: one cache pass + one storage transaction."""
+        hashes = list(hashes)
+        with self._lock:
+            for h in hashes:
+                self._release_cached(h)
+        self._storage.aux_delete_many(AUX_MASK_PREFIX + h for h in hashes)
 
     def get(self, alloc_hash: str) -> Optional[dict]:
         raw = self._storage.aux_get(AUX_MASK_PREFIX + alloc_hash)

@@ -95,6 +95,9 @@ class GPUSharePlugin:
                 log.warning("GC: API check failed for %s/%s: %s", pi.namespace, pi.name, e)
 
         self.cfg.storage.for_each(visit)
+        import os
+
+        aux_keys = []
         for pi in doomed:
             for container, device in pi.container_device_map.items():
                 links = (
@@ -105,18 +108,23 @@ class GPUSharePlugin:
                 for i in range(links):
                     self.cfg.operator.delete(-1, f"{device.hash}-{i}")
                 if self.cfg.cumask:
-                    self.cfg.cumask.release(device.hash)
+                    aux_keys.append("mask/" + device.hash)
                 if self.cfg.limits:
                     self.cfg.limits.delete(device.hash)
                 try:  # hook-recorded pid file (occupancy attribution)
-                    import os
-
                     os.unlink(
                         os.path.join(self.cfg.paths.state_dir, "pids", device.hash)
                     )
                 except OSError:
                     pass
-            self.cfg.storage.delete(pi.namespace, pi.name)
+        # storage work batched: one transaction per GC pass, not per pod
+        # (per-pod autocommit deletes fall behind at high churn)
+        if self.cfg.cumask:
+            self.cfg.cumask.release_many(k[len("mask/"):] for k in aux_keys)
+        else:
+            self.cfg.storage.aux_delete_many(aux_keys)
+        self.cfg.storage.delete_many(pi.key() for pi in doomed)
+        for pi in doomed:
             log.info("GC reclaimed %s/%s", pi.namespace, pi.name)
         return len(doomed)
 

@@ -75,6 +75,18 @@ class Storage:
             self._conn.execute("DELETE FROM pods WHERE key=?", (f"{namespace}/{name}",))
             self._conn.commit()
 
+    def delete_many(self, keys) -> None:
+        """Batch delete (one transaction) — GC reclaiming N pods must not pay
+        N fsyncs (falls behind at high churn otherwise)."""
+        keys = list(keys)
+        if not keys:
+            return
+        with self._lock:
+            self._conn.executemany(
+                "DELETE FROM pods WHERE key=?", [(k,) for k in keys]
+            )
+            self._conn.commit()
+
     def for_each(self, fn: Callable[[PodInfo], None]) -> None:
         with self._lock:
             rows = self._conn.execute("SELECT key, val FROM pods").fetchall()
@@ -99,6 +111,14 @@ class Storage:
     def aux_delete(self, key: str) -> None:
         with self._lock:
             self._conn.execute("DELETE FROM aux WHERE key=?", (key,))
+            self._conn.commit()
+
+    def aux_delete_many(self, keys) -> None:
+        keys = list(keys)
+        if not keys:
+            return
+        with self._lock:
+            self._conn.executemany("DELETE FROM aux WHERE key=?", [(k,) for k in keys])
             self._conn.commit()
 
     def aux_items(self, prefix: str = "") -> list:

@@ -165,8 +165,13 @@ def main():
     time.sleep(1.5)
     h.plugin.core.trigger_refresh()  # unblock the watcher stream
 
+    # drain the full backlog, return freed pages, then measure
+    from elastic_gpu_agent_amd.plugins.aggregate import _malloc_trim
+
+    while h.plugin.gc_once() > 0:
+        pass
+    _malloc_trim()
     rss1 = rss_mb()
-    h.plugin.gc_once()
     h.close()
     growth = rss1 - rss0
     allowed = args.rss_limit_mb + counts["prestart"] * args.rss_per_cycle_bytes / 1e6

@@ -6,6 +6,11 @@ Subcommands:
   masks                   list live CU-mask assignments
   occupancy               per-pod live occupancy (amdsmi join)
   partition IDX [MODE]    get/set SPX|DPX|QPX|CPX compute partition
+  drain IDX [--wait [--repartition MODE]]
+                          stop scheduling onto a GPU (devices re-advertised
+                          Unhealthy); --wait blocks until empty; --repartition
+                          then flips the partition and undrains
+  undrain IDX             return a drained GPU to service
   migrate --from BOLT_DB  import a reference agent's BoltDB state
 
 (GC runs inside the agent daemon — event-driven + 60 s timer; there is no
@@ -89,6 +94,64 @@ def cmd_partition(args) -> int:
     return 0
 
 
+def cmd_drain(args) -> int:
+    from ..drain import (clear_drain, list_drains, live_allocations_on, set_drain,
+                         wait_drained)
+    from ..isolation import LimitsWriter
+    from ..storage import Storage
+
+    st = Storage(args.db)
+    try:
+        if args.list:
+            print(json.dumps(list_drains(st), indent=2))
+            return 0
+        if args.index is None:
+            print("drain: an index is required (or --list)", file=sys.stderr)
+            return 2
+        limits = LimitsWriter(args.limits_dir)
+        set_drain(st, args.index, mode=args.repartition)
+        remaining = live_allocations_on(st, args.index, limits)
+        print(f"gpu {args.index} drained for scheduling "
+              f"({len(remaining)} live allocation(s) remaining)")
+        if not args.wait:
+            for row in remaining:
+                print(f"  waiting on {row['pod']}/{row['container']} ({row['resource']})")
+            if args.repartition:
+                print(f"re-run with --wait to repartition to {args.repartition} "
+                      "once the GPU is empty")
+            return 0
+        ok = wait_drained(
+            st, args.index, limits, timeout=args.timeout,
+            progress=lambda rows: print(f"  {len(rows)} allocation(s) still live..."),
+        )
+        if not ok:
+            print(f"timeout after {args.timeout}s; drain flag left set", file=sys.stderr)
+            return 1
+        print(f"gpu {args.index} is empty")
+        if args.repartition:
+            from elastic_gpu_agent_amd import _amdsmi
+
+            _amdsmi.set_compute_partition(args.index, args.repartition)
+            clear_drain(st, args.index)
+            print(f"gpu {args.index} repartitioned to {args.repartition} and "
+                  "returned to service (agent re-advertises on its next refresh)")
+        return 0
+    finally:
+        st.close()
+
+
+def cmd_undrain(args) -> int:
+    from ..drain import clear_drain
+    from ..storage import Storage
+
+    st = Storage(args.db)
+    clear_drain(st, args.index)
+    st.close()
+    print(f"gpu {args.index} returned to service "
+          "(agent re-advertises on its next refresh)")
+    return 0
+
+
 def cmd_migrate(args) -> int:
     from ..storage import Storage, migrate_from_bolt
 
@@ -115,6 +178,22 @@ def main(argv=None) -> int:
     m = sub.add_parser("migrate")
     m.add_argument("--from", required=True)
     m.set_defaults(fn=cmd_migrate)
+    dr = sub.add_parser(
+        "drain",
+        help="stop scheduling onto a GPU; optionally wait-empty and repartition",
+    )
+    dr.add_argument("index", type=int, nargs="?", default=None)
+    dr.add_argument("--list", action="store_true", help="show active drains")
+    dr.add_argument("--wait", action="store_true",
+                    help="block until no live allocation references the GPU")
+    dr.add_argument("--timeout", type=float, default=600.0)
+    dr.add_argument("--repartition", default=None,
+                    choices=["SPX", "DPX", "QPX", "CPX"],
+                    help="with --wait: set this partition once empty, then undrain")
+    dr.set_defaults(fn=cmd_drain)
+    ud = sub.add_parser("undrain", help="return a drained GPU to service")
+    ud.add_argument("index", type=int)
+    ud.set_defaults(fn=cmd_undrain)
     pt = sub.add_parser("partition", help="get/set SPX|DPX|QPX|CPX compute partition")
     pt.add_argument("index", type=int)
     pt.add_argument("mode", nargs="?", default=None,

@@ -43,10 +43,28 @@ class GPUSharePluginBase:
     def fake_device_ids_for_gpu(self, gpu) -> List[str]:
         raise NotImplementedError
 
+    def _drained(self) -> set:
+        """GPUs under an operator drain (egpuctl drain): advertised Unhealthy
+        so kubelet stops placing pods while existing ones finish."""
+        if self.cfg.storage is None:
+            return set()
+        from ..drain import drained_indexes
+
+        try:
+            return drained_indexes(self.cfg.storage)
+        except Exception as e:  # a broken flag read must not kill ListAndWatch
+            log.error("drain flags unreadable: %s", e)
+            return set()
+
     def list_devices(self, healthy: bool = True) -> List[dict]:
-        health = consts.HEALTHY if healthy else consts.UNHEALTHY
+        drained = self._drained()
         out = []
         for gpu in self.cfg.operator.devices():
+            health = (
+                consts.HEALTHY
+                if healthy and gpu.index not in drained
+                else consts.UNHEALTHY
+            )
             topo = {"nodes": [{"ID": gpu.numa_node}]}
             for did in self.fake_device_ids_for_gpu(gpu):
                 out.append({"ID": did, "health": health, "topology": topo})
@@ -56,9 +74,14 @@ class GPUSharePluginBase:
         """[(ids, encoded Device suffix)] per GPU — the fast-encode shape."""
         from ..protos import fastpath
 
-        health = consts.HEALTHY if healthy else consts.UNHEALTHY
+        drained = self._drained()
         groups = []
         for gpu in self.cfg.operator.devices():
+            health = (
+                consts.HEALTHY
+                if healthy and gpu.index not in drained
+                else consts.UNHEALTHY
+            )
             suffix = fastpath.device_suffix(health, gpu.numa_node)
             groups.append((self.fake_device_ids_for_gpu(gpu), suffix))
         return groups

@@ -112,12 +112,13 @@ def build_kernels(force=False):
 
 
 def build_hook(force=False):
-    src = os.path.join(HERE, "egpu_hook.cpp")
+    srcs = [os.path.join(HERE, "egpu_hook.cpp"), os.path.join(HERE, "devfilter.cpp")]
+    hdr = os.path.join(HERE, "devfilter.h")
     os.makedirs(os.path.join(REPO, "bin"), exist_ok=True)
     out = os.path.join(REPO, "bin", "egpu-hook")
-    if not force and _newer(out, src):
+    if not force and all(_newer(out, s) for s in srcs + [hdr]):
         return out
-    _run(["g++", "-O2", "-std=c++17", src, "-o", out])
+    _run(["g++", "-O2", "-std=c++17", *srcs, "-o", out])
     return out
 
 

@@ -1,0 +1,272 @@
+// devfilter — see devfilter.h. Hand-assembled eBPF (no libbpf dependency;
+// the program is ~6 instructions per rule and the hook must stay a single
+// static binary).
+//
+// Reference parity note: the reference's hook (cmd/elastic-gpu-hook/main.go)
+// predates cgroup v2 and never touches device cgroups at all — it delegates
+// node injection to nvidia-container-cli, which in turn relies on runc
+// having already granted the nodes. This module is the MI355X-native
+// equivalent of that missing layer for v2 hosts.
+
+#include "devfilter.h"
+
+#include <fcntl.h>
+#include <sys/stat.h>
+#include <sys/syscall.h>
+#include <unistd.h>
+
+#include <cerrno>
+#include <cstdlib>
+#include <cstring>
+
+namespace devfilter {
+namespace {
+
+int sys_bpf(int cmd, union bpf_attr* attr, unsigned int size) {
+  return (int)syscall(__NR_bpf, cmd, attr, size);
+}
+
+bpf_insn ins(__u8 code, __u8 dst, __u8 src, __s16 off, __s32 imm) {
+  bpf_insn i{};
+  i.code = code;
+  i.dst_reg = dst;
+  i.src_reg = src;
+  i.off = off;
+  i.imm = imm;
+  return i;
+}
+
+constexpr __u8 R0 = 0, R1 = 1, R2 = 2, R3 = 3, R4 = 4, R5 = 5, R6 = 6;
+
+}  // namespace
+
+std::vector<bpf_insn> build_prog(const std::vector<DevRule>& rules, bool default_allow) {
+  // struct bpf_cgroup_dev_ctx { u32 access_type; u32 major; u32 minor; }
+  // access_type = (access_bits << 16) | dev_type
+  std::vector<bpf_insn> p;
+  // r2 = ctx->access_type; r3 = dev type; r2 = access bits
+  p.push_back(ins(BPF_LDX | BPF_MEM | BPF_W, R2, R1, 0, 0));
+  p.push_back(ins(BPF_ALU64 | BPF_MOV | BPF_X, R3, R2, 0, 0));
+  p.push_back(ins(BPF_ALU64 | BPF_AND | BPF_K, R3, 0, 0, 0xFFFF));
+  p.push_back(ins(BPF_ALU64 | BPF_RSH | BPF_K, R2, 0, 0, 16));
+  // r4 = major; r5 = minor
+  p.push_back(ins(BPF_LDX | BPF_MEM | BPF_W, R4, R1, 4, 0));
+  p.push_back(ins(BPF_LDX | BPF_MEM | BPF_W, R5, R1, 8, 0));
+
+  for (const auto& r : rules) {
+    // emit the rule block with placeholder jump targets, then fix them to
+    // point just past the block
+    std::vector<bpf_insn> b;
+    std::vector<size_t> fixups;  // indexes in b whose off -> block end
+    if (r.type == 'c' || r.type == 'b') {
+      __s32 want = (r.type == 'b') ? BPF_DEVCG_DEV_BLOCK : BPF_DEVCG_DEV_CHAR;
+      fixups.push_back(b.size());
+      b.push_back(ins(BPF_JMP | BPF_JNE | BPF_K, R3, 0, 0, want));
+    }
+    if (r.allow) {
+      // requested ⊆ rule.access: (requested & ~access) != 0 → no match
+      b.push_back(ins(BPF_ALU64 | BPF_MOV | BPF_X, R6, R2, 0, 0));
+      b.push_back(ins(BPF_ALU64 | BPF_AND | BPF_K, R6, 0, 0, (__s32)(~r.access & 7)));
+      fixups.push_back(b.size());
+      b.push_back(ins(BPF_JMP | BPF_JNE | BPF_K, R6, 0, 0, 0));
+    } else {
+      // requested ∩ rule.access empty → no match
+      b.push_back(ins(BPF_ALU64 | BPF_MOV | BPF_X, R6, R2, 0, 0));
+      b.push_back(ins(BPF_ALU64 | BPF_AND | BPF_K, R6, 0, 0, (__s32)(r.access & 7)));
+      fixups.push_back(b.size());
+      b.push_back(ins(BPF_JMP | BPF_JEQ | BPF_K, R6, 0, 0, 0));
+    }
+    if (r.maj >= 0) {
+      fixups.push_back(b.size());
+      b.push_back(ins(BPF_JMP | BPF_JNE | BPF_K, R4, 0, 0, (__s32)r.maj));
+    }
+    if (r.min >= 0) {
+      fixups.push_back(b.size());
+      b.push_back(ins(BPF_JMP | BPF_JNE | BPF_K, R5, 0, 0, (__s32)r.min));
+    }
+    b.push_back(ins(BPF_ALU64 | BPF_MOV | BPF_K, R0, 0, 0, r.allow ? 1 : 0));
+    b.push_back(ins(BPF_JMP | BPF_EXIT, 0, 0, 0, 0));
+    for (size_t idx : fixups) b[idx].off = (__s16)(b.size() - idx - 1);
+    p.insert(p.end(), b.begin(), b.end());
+  }
+
+  p.push_back(ins(BPF_ALU64 | BPF_MOV | BPF_K, R0, 0, 0, default_allow ? 1 : 0));
+  p.push_back(ins(BPF_JMP | BPF_EXIT, 0, 0, 0, 0));
+  return p;
+}
+
+int load_prog(const std::vector<bpf_insn>& insns, std::string* err) {
+  union bpf_attr attr;
+  memset(&attr, 0, sizeof(attr));
+  attr.prog_type = BPF_PROG_TYPE_CGROUP_DEVICE;
+  attr.insn_cnt = (__u32)insns.size();
+  attr.insns = (__u64)(uintptr_t)insns.data();
+  // no GPL-only helpers are used, so any license string loads
+  static const char license[] = "Apache-2.0";
+  attr.license = (__u64)(uintptr_t)license;
+  int fd = sys_bpf(BPF_PROG_LOAD, &attr, sizeof(attr));
+  if (fd < 0 && err) *err = std::string("BPF_PROG_LOAD: ") + strerror(errno);
+  return fd;
+}
+
+namespace {
+
+int query_ids(int cg_fd, std::vector<__u32>* ids, std::string* err) {
+  union bpf_attr attr;
+  memset(&attr, 0, sizeof(attr));
+  attr.query.target_fd = cg_fd;
+  attr.query.attach_type = BPF_CGROUP_DEVICE;
+  ids->resize(64);
+  attr.query.prog_ids = (__u64)(uintptr_t)ids->data();
+  attr.query.prog_cnt = (__u32)ids->size();
+  if (sys_bpf(BPF_PROG_QUERY, &attr, sizeof(attr)) != 0) {
+    if (err) *err = std::string("BPF_PROG_QUERY: ") + strerror(errno);
+    return -1;
+  }
+  ids->resize(attr.query.prog_cnt);
+  return 0;
+}
+
+}  // namespace
+
+int query_attached_count(const std::string& cgroup_dir, std::string* err) {
+  int cg = open(cgroup_dir.c_str(), O_RDONLY | O_DIRECTORY);
+  if (cg < 0) {
+    if (err) *err = "open " + cgroup_dir + ": " + strerror(errno);
+    return -1;
+  }
+  std::vector<__u32> ids;
+  int rc = query_ids(cg, &ids, err);
+  close(cg);
+  return rc == 0 ? (int)ids.size() : -1;
+}
+
+int replace_attached(const std::string& cgroup_dir, int prog_fd, std::string* err) {
+  int cg = open(cgroup_dir.c_str(), O_RDONLY | O_DIRECTORY);
+  if (cg < 0) {
+    if (err) *err = "open " + cgroup_dir + ": " + strerror(errno);
+    return -1;
+  }
+  std::vector<__u32> old_ids;
+  if (query_ids(cg, &old_ids, err) != 0) {
+    close(cg);
+    return -1;
+  }
+  // attach the union program FIRST so there is no window with the grant
+  // absent; AND-semantics make the overlap window merely conservative
+  union bpf_attr attr;
+  memset(&attr, 0, sizeof(attr));
+  attr.target_fd = cg;
+  attr.attach_bpf_fd = prog_fd;
+  attr.attach_type = BPF_CGROUP_DEVICE;
+  attr.attach_flags = BPF_F_ALLOW_MULTI;
+  if (sys_bpf(BPF_PROG_ATTACH, &attr, sizeof(attr)) != 0) {
+    if (err) *err = std::string("BPF_PROG_ATTACH: ") + strerror(errno);
+    close(cg);
+    return -1;
+  }
+  int rc = 0;
+  for (__u32 id : old_ids) {
+    union bpf_attr get;
+    memset(&get, 0, sizeof(get));
+    get.prog_id = id;
+    int old_fd = sys_bpf(BPF_PROG_GET_FD_BY_ID, &get, sizeof(get));
+    if (old_fd < 0) {
+      if (err) *err = std::string("BPF_PROG_GET_FD_BY_ID: ") + strerror(errno);
+      rc = -1;
+      continue;
+    }
+    union bpf_attr det;
+    memset(&det, 0, sizeof(det));
+    det.target_fd = cg;
+    det.attach_bpf_fd = old_fd;
+    det.attach_type = BPF_CGROUP_DEVICE;
+    if (sys_bpf(BPF_PROG_DETACH, &det, sizeof(det)) != 0) {
+      if (err) *err = std::string("BPF_PROG_DETACH: ") + strerror(errno);
+      rc = -1;
+    }
+    close(old_fd);
+  }
+  close(cg);
+  return rc;
+}
+
+std::vector<DevRule> parse_oci_device_rules(const std::string& config, bool* found) {
+  std::vector<DevRule> rules;
+  if (found) *found = false;
+  size_t res = config.find("\"resources\"");
+  if (res == std::string::npos) return rules;
+  size_t dev = config.find("\"devices\"", res);
+  if (dev == std::string::npos) return rules;
+  size_t open_b = config.find('[', dev);
+  if (open_b == std::string::npos) return rules;
+  size_t close_b = config.find(']', open_b);  // array of flat objects
+  if (close_b == std::string::npos) return rules;
+  if (found) *found = true;
+  size_t pos = open_b;
+  while (true) {
+    size_t ob = config.find('{', pos);
+    if (ob == std::string::npos || ob > close_b) break;
+    size_t cb = config.find('}', ob);
+    if (cb == std::string::npos || cb > close_b) break;
+    std::string obj = config.substr(ob, cb - ob + 1);
+    DevRule r;
+    r.allow = obj.find("\"allow\":true") != std::string::npos ||
+              obj.find("\"allow\": true") != std::string::npos;
+    size_t t = obj.find("\"type\"");
+    if (t != std::string::npos) {
+      size_t q = obj.find('"', obj.find(':', t));
+      if (q != std::string::npos && q + 1 < obj.size()) r.type = obj[q + 1];
+    }
+    auto find_ll = [&obj](const char* key, long long* out) {
+      size_t k = obj.find(key);
+      if (k == std::string::npos) return;
+      size_t c = obj.find(':', k);
+      if (c == std::string::npos) return;
+      *out = strtoll(obj.c_str() + c + 1, nullptr, 10);
+    };
+    find_ll("\"major\"", &r.maj);
+    find_ll("\"minor\"", &r.min);
+    size_t a = obj.find("\"access\"");
+    if (a != std::string::npos) {
+      size_t q = obj.find('"', obj.find(':', a));
+      if (q != std::string::npos) {
+        r.access = 0;
+        for (size_t i = q + 1; i < obj.size() && obj[i] != '"'; ++i) {
+          if (obj[i] == 'r') r.access |= BPF_DEVCG_ACC_READ;
+          if (obj[i] == 'w') r.access |= BPF_DEVCG_ACC_WRITE;
+          if (obj[i] == 'm') r.access |= BPF_DEVCG_ACC_MKNOD;
+        }
+      }
+    }
+    rules.push_back(r);
+    pos = cb + 1;
+  }
+  return rules;
+}
+
+std::string unified_cgroup_dir(long pid, const std::string& cgroup_root) {
+  std::string path = "/proc/" + std::to_string(pid) + "/cgroup";
+  FILE* f = fopen(path.c_str(), "r");
+  if (!f) return {};
+  char line[4096];
+  std::string out;
+  while (fgets(line, sizeof(line), f)) {
+    if (strncmp(line, "0::", 3) == 0) {
+      std::string p = line + 3;
+      while (!p.empty() && (p.back() == '\n' || p.back() == '\r')) p.pop_back();
+      out = cgroup_root + p;
+      break;
+    }
+  }
+  fclose(f);
+  return out;
+}
+
+bool host_is_pure_v2(const std::string& cgroup_root) {
+  struct stat st {};
+  if (stat((cgroup_root + "/cgroup.controllers").c_str(), &st) != 0) return false;
+  return stat((cgroup_root + "/devices").c_str(), &st) != 0;
+}
+
+}  // namespace devfilter

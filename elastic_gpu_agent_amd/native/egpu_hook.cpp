@@ -14,8 +14,12 @@
 //      agent's PreStartContainer) → /dev/dri/renderD<minor> targets;
 //   4. enter the container's mount namespace (setns) and mknod
 //      /dev/kfd + /dev/dri/renderD<minor> with the host major:minor;
-//   5. for cgroup-v1 hosts, whitelist the nodes in the devices cgroup
-//      (v2 grants flow through the kubelet DeviceSpec path instead).
+//   5. grant the nodes in the device cgroup: on v1 hosts via
+//      devices.allow; on pure-v2 hosts by replacing the container's
+//      BPF_PROG_TYPE_CGROUP_DEVICE program with one whose allowlist is
+//      the union of the container's OCI rules and the GPU nodes
+//      (devfilter.cpp — multi-attach cannot widen access, the kernel
+//      ANDs all attached programs' verdicts).
 //
 // No GPU env → passthrough (exit 0), like the reference hook.
 //
@@ -37,6 +41,8 @@
 #include <cstring>
 #include <string>
 #include <vector>
+
+#include "devfilter.h"
 
 namespace {
 
@@ -191,7 +197,60 @@ std::string devices_cgroup_path(long pid) {
   return {};
 }
 
-int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun) {
+// pure-v2 hosts: replace the container's device-filter program with one
+// that also allows the GPU nodes. Failures are logged, not fatal — the
+// kubelet DeviceSpec path normally granted the nodes already; this makes
+// the grant independent of that resolution order (ROADMAP item 1).
+void grant_v2(long pid, const std::vector<DeviceNode>& nodes, const std::string& config) {
+  std::string cgdir = devfilter::unified_cgroup_dir(pid);
+  if (cgdir.empty()) {
+    logf("WARN v2: pid %ld has no unified-cgroup line", pid);
+    return;
+  }
+  std::string err;
+  int count = devfilter::query_attached_count(cgdir, &err);
+  if (count < 0) {
+    logf("WARN v2: %s", err.c_str());
+    return;
+  }
+  if (count == 0) {
+    // no filter attached ⇒ device access already unrestricted; attaching
+    // an allowlist here would REMOVE access, so do nothing
+    logf("v2: no device filter on %s; nothing to replace", cgdir.c_str());
+    return;
+  }
+  bool found = false;
+  auto oci = devfilter::parse_oci_device_rules(config, &found);
+  std::vector<devfilter::DevRule> rules;
+  for (const auto& n : nodes) {
+    devfilter::DevRule r;
+    r.type = 'c';
+    r.maj = n.maj;
+    r.min = n.min;
+    r.access = 7;
+    r.allow = true;
+    rules.push_back(r);
+  }
+  // the OCI list has sequential (last-write-wins) semantics; the program is
+  // first-match, so append the container's rules in reverse order
+  for (auto it = oci.rbegin(); it != oci.rend(); ++it) rules.push_back(*it);
+  auto prog = devfilter::build_prog(rules, /*default_allow=*/false);
+  int fd = devfilter::load_prog(prog, &err);
+  if (fd < 0) {
+    logf("WARN v2: %s", err.c_str());
+    return;
+  }
+  if (devfilter::replace_attached(cgdir, fd, &err) != 0) {
+    logf("WARN v2: %s", err.c_str());
+  } else {
+    logf("v2: device filter replaced on %s (%zu rules, %zu from OCI config)",
+         cgdir.c_str(), rules.size(), oci.size());
+  }
+  close(fd);
+}
+
+int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun,
+           const std::string& config) {
   if (dryrun) {
     printf("{\"pid\": %ld, \"nodes\": [", pid);
     for (size_t i = 0; i < nodes.size(); ++i) {
@@ -202,7 +261,7 @@ int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun) {
     return 0;
   }
 
-  // cgroup v1 allow-list (v2 device access was granted via kubelet DeviceSpec)
+  // device-cgroup grant: v1 devices.allow, or v2 eBPF filter replacement
   std::string cg = devices_cgroup_path(pid);
   if (!cg.empty()) {
     std::string allow = cg + "/devices.allow";
@@ -217,6 +276,8 @@ int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun) {
     } else {
       logf("WARN: cannot open %s: %s", allow.c_str(), strerror(errno));
     }
+  } else if (devfilter::host_is_pure_v2()) {
+    grant_v2(pid, nodes, config);
   }
 
   // enter the container's mount namespace and create the nodes
@@ -257,7 +318,43 @@ int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun) {
 
 }  // namespace
 
+// self-test subcommands (exercised by tests/test_devfilter.py against the
+// real kernel verifier / a scratch cgroup):
+//   egpu-hook devfilter-load <config.json>
+//   egpu-hook devfilter-attach <cgroup_dir> <config.json>
+int devfilter_cmd(int argc, char** argv) {
+  bool do_attach = strcmp(argv[1], "devfilter-attach") == 0;
+  const char* cfg_path = do_attach ? argv[3] : argv[2];
+  std::string config = slurp_file(cfg_path);
+  if (config.empty()) {
+    fprintf(stderr, "cannot read %s\n", cfg_path);
+    return 1;
+  }
+  bool found = false;
+  auto oci = devfilter::parse_oci_device_rules(config, &found);
+  std::vector<devfilter::DevRule> rules(oci.rbegin(), oci.rend());
+  auto prog = devfilter::build_prog(rules, /*default_allow=*/false);
+  std::string err;
+  int fd = devfilter::load_prog(prog, &err);
+  if (fd < 0) {
+    fprintf(stderr, "%s\n", err.c_str());
+    return 1;
+  }
+  if (do_attach && devfilter::replace_attached(argv[2], fd, &err) != 0) {
+    fprintf(stderr, "%s\n", err.c_str());
+    close(fd);
+    return 1;
+  }
+  printf("ok insns=%zu rules=%zu found=%d\n", prog.size(), rules.size(), found ? 1 : 0);
+  close(fd);
+  return 0;
+}
+
 int main(int argc, char** argv) {
+  if (argc > 2 && (strcmp(argv[1], "devfilter-load") == 0 ||
+                   (argc > 3 && strcmp(argv[1], "devfilter-attach") == 0))) {
+    return devfilter_cmd(argc, argv);
+  }
   // accept NVIDIA-hook-style lifecycle argument; only prestart acts
   if (argc > 1 && strcmp(argv[1], "prestart") != 0 && strcmp(argv[1], "createRuntime") != 0) {
     return 0;
@@ -351,5 +448,5 @@ int main(int argc, char** argv) {
       fclose(pf);
     }
   }
-  return inject(pid, nodes, dryrun);
+  return inject(pid, nodes, dryrun, config);
 }

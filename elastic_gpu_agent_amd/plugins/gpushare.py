@@ -150,11 +150,19 @@ class GPUSharePluginBase:
         devices = self.cfg.operator.devices()
         responses = []
         for cr in request.get("container_requests", []):
+            size = cr.get("allocation_size", 0)
+            # memory allocations and fractional core requests bind to exactly
+            # one physical GPU at PreStart — never prefer a spanning set
+            single = (
+                self.resource_name == consts.RESOURCE_GPU_MEMORY
+                or size <= consts.GPU_PERCENT_EACH_CARD
+            )
             picked = topology.prefer_allocation(
                 cr.get("available_deviceIDs", []),
                 cr.get("must_include_deviceIDs", []),
-                cr.get("allocation_size", 0),
+                size,
                 devices,
+                single_gpu=single,
             )
             responses.append({"deviceIDs": picked})
         return {"container_responses": responses}

@@ -35,9 +35,22 @@ class SimScheduler:
             for slot in range(gpu.memory_mib // self.mem_unit_mib):
                 self.free_mem[f"{gpu.index}-{slot:06d}"] = True
 
+    # kubelet removes Unhealthy devices from the allocatable pool; mirror
+    # that so drains (per-GPU Unhealthy re-advertisement) stop placements
+    unhealthy: set = field(default_factory=set)
+
     def _available(self, resource: str) -> List[str]:
         pool = self.free_core if resource == consts.RESOURCE_GPU_CORE else self.free_mem
-        return [k for k, free in pool.items() if free]
+        return [k for k, free in pool.items() if free and k not in self.unhealthy]
+
+    def sync_health(self, *plugins) -> None:
+        """Ingest ListAndWatch health from agent plugin(s), as kubelet would."""
+        for plugin in plugins:
+            for dev in plugin.list_devices(True):
+                if dev["health"] == consts.HEALTHY:
+                    self.unhealthy.discard(dev["ID"])
+                else:
+                    self.unhealthy.add(dev["ID"])
 
     def place(
         self,
@@ -64,7 +77,10 @@ class SimScheduler:
             if preferred_fn is not None:
                 ids = preferred_fn(resource, avail, size)
             else:
-                ids = prefer_allocation(avail, [], size, self.devices)
+                single = (resource == consts.RESOURCE_GPU_MEMORY
+                          or size <= consts.GPU_PERCENT_EACH_CARD)
+                ids = prefer_allocation(avail, [], size, self.devices,
+                                        single_gpu=single)
             return ids if len(ids) == size else None
 
         if core_units:

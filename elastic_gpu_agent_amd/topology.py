@@ -44,6 +44,7 @@ def prefer_allocation(
     must_include: Sequence[str],
     size: int,
     devices: List[GPUDevice],
+    single_gpu: bool = False,
 ) -> List[str]:
     """Pick ``size`` device IDs from ``available`` ⊇ ``must_include``.
 
@@ -53,6 +54,12 @@ def prefer_allocation(
        fewest-fragment GPUs (pack fractions tightly → whole GPUs stay free);
     3. when several GPUs are needed, choose the set maximizing pairwise xGMI
        links, tie-broken by NUMA-node co-location and lower index.
+
+    ``single_gpu=True`` restricts the pick to ONE GPU (required for every
+    memory allocation and for fractional core requests — PreStart binds
+    those to exactly one physical GPU); when no single GPU has ``size``
+    units free the pick is truncated, which callers treat as "does not
+    fit" rather than silently producing an unbindable spanning set.
     """
     dev_by_idx = {d.index: d for d in devices}
     chosen: List[str] = list(must_include)
@@ -92,6 +99,20 @@ def prefer_allocation(
 
         idxs.sort(key=key)
         return idxs
+
+    if single_gpu:
+        # all units must come from one GPU (the one must_include touched,
+        # else the best-ranked GPU with enough room)
+        if touched:
+            g = next(iter(touched))
+            pool = groups.get(g, [])
+            chosen.extend(pool[:remaining])
+            return chosen
+        for g in gpu_order():
+            if len(groups[g]) >= remaining:
+                chosen.extend(groups[g][:remaining])
+                return chosen
+        return chosen  # nothing fits: short pick, caller rejects
 
     while remaining > 0 and groups:
         order = gpu_order()

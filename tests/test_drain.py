@@ -178,6 +178,31 @@ def test_drain_cli_wait_timeout(tmp_path):
         h.close()
 
 
+def test_drain_stops_simulated_scheduling(tmp_path):
+    """End-to-end with the scheduler sim: after a drain + health sync (what
+    kubelet does with the Unhealthy re-advertisement), new pods land on the
+    other GPU only; undrain + sync restores placement."""
+    from elastic_gpu_agent_amd.schedsim import SimScheduler
+
+    h = Harness(str(tmp_path), gpus=2)
+    try:
+        sim = SimScheduler(devices=h.operator.devices(), mem_unit_mib=1024)
+        set_drain(h.storage, 0)
+        sim.sync_health(h.plugin.core, h.plugin.memory)
+        for i in range(3):
+            placed = sim.place(f"c{i}", core_units=30)
+            assert placed is not None
+            assert placed["gpu_indexes"] == [1], placed
+        # GPU 1 has 10 core units left; a 30-unit pod no longer fits anywhere
+        assert sim.place("cx", core_units=30) is None
+        clear_drain(h.storage, 0)
+        sim.sync_health(h.plugin.core, h.plugin.memory)
+        placed = sim.place("cy", core_units=30)
+        assert placed is not None and placed["gpu_indexes"] == [0]
+    finally:
+        h.close()
+
+
 def test_drain_cli_requires_index(tmp_path):
     h = Harness(str(tmp_path), gpus=1)
     try:

@@ -43,3 +43,27 @@ def test_xgmi_score_full_mesh():
     d = {g.index: g for g in devs(8)}
     assert xgmi_score([0, 1, 2, 3], d) == 6  # all pairs linked on a full mesh
     assert xgmi_score([0], d) == 0
+
+
+def test_single_gpu_never_spans():
+    """Fractional requests must come from ONE GPU even when packing would
+    prefer spreading over fragmented ones (a spanning fractional set is
+    unbindable at PreStart — found via the drain/schedsim test)."""
+    # GPU 0 has only 20 free, GPU 1 has 40 free
+    avail = [f"0-{i:02d}" for i in range(20)] + [f"1-{i:02d}" for i in range(40)]
+    picked = prefer_allocation(avail, [], 30, devs(2), single_gpu=True)
+    assert len(picked) == 30
+    assert all(d.startswith("1-") for d in picked)
+
+
+def test_single_gpu_short_pick_when_nothing_fits():
+    avail = [f"0-{i:02d}" for i in range(20)] + [f"1-{i:02d}" for i in range(25)]
+    picked = prefer_allocation(avail, [], 30, devs(2), single_gpu=True)
+    assert len(picked) < 30  # caller treats as does-not-fit
+
+
+def test_single_gpu_follows_must_include():
+    avail = [f"{g}-{i:02d}" for g in range(2) for i in range(50)]
+    picked = prefer_allocation(avail, ["1-07"], 10, devs(2), single_gpu=True)
+    assert len(picked) == 10
+    assert all(d.startswith("1-") for d in picked)

@@ -203,6 +203,67 @@ def test_drain_stops_simulated_scheduling(tmp_path):
         h.close()
 
 
+def test_drain_wait_repartition_cycle(tmp_path, monkeypatch):
+    """Full lifecycle: drain → GPU empties → partition flipped via amdsmi →
+    flag cleared (GPU back in service)."""
+    import elastic_gpu_agent_amd as pkg
+
+    calls = []
+
+    class StubSmi:
+        @staticmethod
+        def set_compute_partition(index, mode):
+            calls.append((index, mode))
+
+        @staticmethod
+        def get_compute_partition(index):
+            return calls[-1][1] if calls else "SPX"
+
+    monkeypatch.setattr(pkg, "_amdsmi", StubSmi, raising=False)
+    h = Harness(str(tmp_path), gpus=2)
+    try:
+        _bind_fractional(h, "ns", "p0", "c", 0, 25)
+
+        import threading
+
+        def finish_pod():
+            import time
+
+            time.sleep(0.15)
+            h.sitter.remove("ns", "p0")
+            h.plugin.gc_once()
+
+        t = threading.Thread(target=finish_pod)
+        t.start()
+        rc = egpuctl.main(
+            ["--db", h.storage.path, "--limits-dir", str(tmp_path / "limits"),
+             "drain", "0", "--wait", "--timeout", "5", "--repartition", "CPX"]
+        )
+        t.join()
+        assert rc == 0
+        assert calls == [(0, "CPX")]
+        assert drained_indexes(h.storage) == set()  # undrained after flip
+    finally:
+        h.close()
+
+
+def test_drain_survives_restart(tmp_path):
+    """The drain flag lives in the persisted aux table: a restarted agent
+    (fresh plugin over the same DB) still advertises the GPU Unhealthy."""
+    h = Harness(str(tmp_path), gpus=2)
+    try:
+        set_drain(h.storage, 0)
+    finally:
+        h.close()
+    h2 = Harness(str(tmp_path), gpus=2)
+    try:
+        by_gpu = _health_by_gpu(h2.plugin.core.list_devices(True))
+        assert by_gpu[0] == {consts.UNHEALTHY}
+        assert by_gpu[1] == {consts.HEALTHY}
+    finally:
+        h2.close()
+
+
 def test_drain_cli_requires_index(tmp_path):
     h = Harness(str(tmp_path), gpus=1)
     try:

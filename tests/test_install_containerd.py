@@ -154,3 +154,24 @@ def test_rejects_config_without_runtime(tmp_path):
                   "--hook", "/usr/local/bin/egpu-hook"])
     assert rc == 1
     assert cfg.read_text() == "version = 2\n"
+
+
+def test_host_root_split_view(tmp_path):
+    """Init-container scenario: the TOML records the HOST-view spec path
+    while the file is written through the /host mount (the agent-view vs.
+    advertise-view split that bit the limits mounts too)."""
+    host = tmp_path / "host"
+    (host / "etc" / "containerd").mkdir(parents=True)
+    cfg = host / "etc" / "containerd" / "config.toml"
+    cfg.write_text(CONFIG_1X)
+    rc = ic.main(["--config", str(cfg),
+                  "--spec", "/etc/containerd/egpu-base.json",
+                  "--host-root", str(host),
+                  "--hook", "/usr/local/bin/egpu-hook"])
+    assert rc == 0
+    parsed = tomli.loads(cfg.read_text())
+    # TOML carries the host-view path verbatim
+    assert ic.existing_base_spec(parsed) == "/etc/containerd/egpu-base.json"
+    # ... but the file landed under the mount
+    body = json.loads((host / "etc" / "containerd" / "egpu-base.json").read_text())
+    assert body["hooks"]["prestart"][0]["path"] == "/usr/local/bin/egpu-hook"

@@ -14,8 +14,19 @@ chmod 0755 /host/usr/local/bin/egpu-hook
 mkdir -p /host/opt/egpu
 cp /opt/agent/elastic_gpu_agent_amd/libegpu_shim.so /host/opt/egpu/libegpu_shim.so
 
-# 3. OCI hooks.d registration (CRI-O / podman style). containerd users add
-#    the hook via the runtime handler config instead; see docs/DEPLOY.md.
+# 3. containerd registration (base_runtime_spec), when a containerd config
+#    is visible on the host mount. Idempotent; set EGPU_SKIP_CONTAINERD=1 to
+#    opt out. The node owner restarts containerd to activate.
+if [ -z "$EGPU_SKIP_CONTAINERD" ] && [ -f /host/etc/containerd/config.toml ]; then
+  python3 /opt/agent/tools/install_containerd.py \
+    --config /host/etc/containerd/config.toml \
+    --spec /etc/containerd/egpu-base.json --host-root /host \
+    --hook /usr/local/bin/egpu-hook || \
+    echo "WARN: containerd registration failed (see docs/DEPLOY.md)"
+fi
+
+# 4. OCI hooks.d registration (CRI-O / podman style). containerd users get
+#    the base_runtime_spec registration above instead; see docs/DEPLOY.md.
 mkdir -p /host/etc/containers/oci/hooks.d
 cat > /host/etc/containers/oci/hooks.d/10-egpu.json <<'EOF'
 {

@@ -172,7 +172,11 @@ def patch_config_text(text: str, spec_path: str, runtime: str = "runc") -> str:
 
 
 def install(config_path: str, spec_path: str, hook_path: str, runtime: str,
-            dry_run: bool) -> int:
+            dry_run: bool, host_root: str = "") -> int:
+    # Paths INSIDE the TOML / spec are host-view (containerd resolves them on
+    # the host); when running from the agent container, file access goes
+    # through the /host mount instead — host_root bridges the two (the same
+    # agent-view vs. advertise-view split as AgentPaths.limits_host_view).
     with open(config_path) as f:
         text = f.read()
     config = _parse_toml(text)
@@ -181,9 +185,10 @@ def install(config_path: str, spec_path: str, hook_path: str, runtime: str,
     patch_toml = target_spec is None
     if target_spec is None:
         target_spec = spec_path
+    spec_file = host_root + target_spec
 
-    if os.path.exists(target_spec):
-        with open(target_spec) as f:
+    if os.path.exists(spec_file):
+        with open(spec_file) as f:
             spec = json.load(f)
     else:
         spec = default_spec()
@@ -206,13 +211,13 @@ def install(config_path: str, spec_path: str, hook_path: str, runtime: str,
         return 0
 
     if spec_changed:
-        tmp = target_spec + ".egpu-tmp"
+        tmp = spec_file + ".egpu-tmp"
         with open(tmp, "w") as f:
             json.dump(spec, f, indent=2)
-        os.replace(tmp, target_spec)
-        print(f"wrote {target_spec} (prestart hook: {hook_path})")
+        os.replace(tmp, spec_file)
+        print(f"wrote {spec_file} (prestart hook: {hook_path})")
     else:
-        print(f"{target_spec}: hook already registered")
+        print(f"{spec_file}: hook already registered")
 
     if patch_toml:
         shutil.copyfile(config_path, config_path + ".egpu-bak")
@@ -236,14 +241,18 @@ def main(argv=None) -> int:
     p = argparse.ArgumentParser(description=__doc__.splitlines()[0])
     p.add_argument("--config", default="/etc/containerd/config.toml")
     p.add_argument("--spec", default="/etc/containerd/egpu-base.json",
-                   help="base spec to create when none is configured")
+                   help="base spec to create when none is configured "
+                        "(HOST-view path — what goes into the TOML)")
     p.add_argument("--hook", default=HOOK_DEFAULT)
     p.add_argument("--runtime", default="runc")
+    p.add_argument("--host-root", default="",
+                   help="prefix for FILE access to host-view paths when "
+                        "running inside the agent container (e.g. /host)")
     p.add_argument("--dry-run", action="store_true")
     args = p.parse_args(argv)
     try:
         return install(args.config, args.spec, args.hook, args.runtime,
-                       args.dry_run)
+                       args.dry_run, host_root=args.host_root)
     except (OSError, ValueError, RuntimeError) as e:
         print(f"error: {e}", file=sys.stderr)
         return 1

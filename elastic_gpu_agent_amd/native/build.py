@@ -63,7 +63,7 @@ def build_fastwire(force=False):
     _run(
         ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src]
         + pybind_includes()
-        + ["-o", out]
+        + ["-o", out, "-lcrypto"]
     )
     return out
 

@@ -20,6 +20,9 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <openssl/evp.h>
+
+#include <algorithm>
 #include <cstdint>
 #include <cstring>
 #include <stdexcept>
@@ -196,6 +199,63 @@ py::bytes encode_nested_string_lists(const std::vector<std::vector<std::string>>
   return py::bytes(out);
 }
 
+// ---- Allocate request digest (hash + count without materializing IDs) ----
+// The reference-exact gpu-memory contract (1-MiB units) makes Allocate
+// requests carry up to ~295k device IDs. The Allocate handler only needs the
+// device-set HASH (sha256 of ":".join(sorted ids), first 8 hex chars —
+// types.hash_device_ids) and the COUNT, so this computes both straight off
+// the wire: no 73k-element Python lists, no Python sort/join/sha. Byte-wise
+// span comparison matches Python's code-point string order because UTF-8 is
+// order-preserving. GIL released during parse/sort/hash.
+
+std::vector<std::pair<std::string, size_t>> digest_raw(const uint8_t* p, size_t len) {
+  std::vector<std::pair<const uint8_t*, size_t>> outer;
+  field1_spans(p, p + len, outer);
+  std::vector<std::pair<std::string, size_t>> result;
+  result.reserve(outer.size());
+  for (auto& o : outer) {
+    std::vector<std::pair<const uint8_t*, size_t>> ids;
+    field1_spans(o.first, o.first + o.second, ids);
+    auto cmp = [](const std::pair<const uint8_t*, size_t>& a,
+                  const std::pair<const uint8_t*, size_t>& b) {
+      int c = memcmp(a.first, b.first, std::min(a.second, b.second));
+      if (c) return c < 0;
+      return a.second < b.second;
+    };
+    if (!std::is_sorted(ids.begin(), ids.end(), cmp))
+      std::sort(ids.begin(), ids.end(), cmp);
+    EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+    EVP_DigestInit_ex(ctx, EVP_sha256(), nullptr);
+    for (size_t i = 0; i < ids.size(); ++i) {
+      if (i) EVP_DigestUpdate(ctx, ":", 1);
+      EVP_DigestUpdate(ctx, ids[i].first, ids[i].second);
+    }
+    unsigned char md[32];
+    unsigned int mdlen = 0;
+    EVP_DigestFinal_ex(ctx, md, &mdlen);
+    EVP_MD_CTX_free(ctx);
+    char hex[9];
+    snprintf(hex, sizeof hex, "%02x%02x%02x%02x", md[0], md[1], md[2], md[3]);
+    result.emplace_back(std::string(hex, 8), ids.size());
+  }
+  return result;
+}
+
+py::list digest_allocate_request(py::bytes data) {
+  char* buf;
+  Py_ssize_t len;
+  PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+  std::vector<std::pair<std::string, size_t>> result;
+  {
+    py::gil_scoped_release rel;
+    result = digest_raw((const uint8_t*)buf, (size_t)len);
+  }
+  py::list out;
+  for (auto& r : result)
+    out.append(py::make_tuple(py::str(r.first), (long long)r.second));
+  return out;
+}
+
 // ---- AllocateResponse encoder (the Allocate hot path's response half) ----
 // Byte-identical to protos.deviceplugin.AllocateResponse.encode (MessageSpec
 // policy: non-repeated strings/bools omitted when empty/false; map entries
@@ -286,4 +346,5 @@ PYBIND11_MODULE(_fastwire, m) {
   m.def("encode_string_list", &encode_string_list);
   m.def("encode_nested_string_lists", &encode_nested_string_lists);
   m.def("encode_allocate_response", &encode_allocate_response);
+  m.def("digest_allocate_request", &digest_allocate_request);
 }

@@ -95,7 +95,7 @@ class DevicePluginServer:
                 "GetPreferredAllocation", p.get_preferred_allocation,
                 dp.PreferredAllocationRequest, dp.PreferredAllocationResponse),
             "Allocate": timed("Allocate", p.allocate,
-                              _FastSpec(fastpath.decode_allocate_request),
+                              _FastSpec(fastpath.decode_allocate_request_digest),
                               _FastSpec(encode=fastpath.encode_allocate_response)),
             "PreStartContainer": timed(
                 "PreStartContainer", p.pre_start_container,

@@ -170,15 +170,20 @@ class GPUSharePluginBase:
     def allocate(self, request, context) -> dict:
         responses = []
         for cr in request.get("container_requests", []):
-            ids = cr.get("devicesIDs", [])
-            device = Device.new(ids, self.resource_name)
-            responses.append(self._allocate_one(device, ids))
+            # server fast path supplies (hash, count) straight off the wire
+            # (fastpath.decode_allocate_request_digest); explicit ID lists
+            # (tests, in-process callers) are hashed here instead
+            digest = cr.get("digest")
+            if digest is None:
+                ids = cr.get("devicesIDs", [])
+                digest = (Device.new(ids, self.resource_name).hash, len(ids))
+            responses.append(self._allocate_one(digest[0], digest[1]))
         return {"container_responses": responses}
 
-    def _allocate_one(self, device: Device, ids: List[str]) -> dict:
+    def _allocate_one(self, alloc_hash: str, n_units: int) -> dict:
         raise NotImplementedError
 
-    def _isolation_payload(self, device: Device, kind: str) -> dict:
+    def _isolation_payload(self, alloc_hash: str, kind: str) -> dict:
         """Common shim env + mounts for a fractional allocation."""
         paths = self.cfg.paths
         resp: dict = {"envs": {}, "mounts": []}
@@ -192,7 +197,7 @@ class GPUSharePluginBase:
         import os as _os
 
         limits_host = paths.limits_host_view(
-            _os.path.basename(self.cfg.limits.host_path(device.hash)))
+            _os.path.basename(self.cfg.limits.host_path(alloc_hash)))
         resp["mounts"] = [
             {
                 "container_path": paths.shim_container_path,
@@ -292,14 +297,14 @@ class GPUShareCorePlugin(GPUSharePluginBase):
         """Number of per-allocation GPU links: one per started 100 units."""
         return max(1, math.ceil(ids_count / consts.GPU_PERCENT_EACH_CARD))
 
-    def _allocate_one(self, device: Device, ids: List[str]) -> dict:
-        n_links = self.links_for(len(ids))
+    def _allocate_one(self, alloc_hash: str, n_units: int) -> dict:
+        n_links = self.links_for(n_units)
         devices_spec = [
             {"container_path": consts.KFD_PATH, "host_path": consts.KFD_PATH,
              "permissions": "rw"}
         ]
         for i in range(n_links):
-            host = f"/dev/{consts.ELASTIC_GPU_LINK_FMT % (device.hash + '-' + str(i))}"
+            host = f"/dev/{consts.ELASTIC_GPU_LINK_FMT % (alloc_hash + '-' + str(i))}"
             devices_spec.append(
                 {
                     "container_path": f"/dev/egpu/gpu{i}",
@@ -308,12 +313,12 @@ class GPUShareCorePlugin(GPUSharePluginBase):
                 }
             )
         resp = {
-            "envs": {consts.GPU_ENV_KEY: device.hash},
+            "envs": {consts.GPU_ENV_KEY: alloc_hash},
             "devices": devices_spec,
         }
-        fractional = len(ids) < consts.GPU_PERCENT_EACH_CARD
+        fractional = n_units < consts.GPU_PERCENT_EACH_CARD
         if fractional:
-            iso = self._isolation_payload(device, "core")
+            iso = self._isolation_payload(alloc_hash, "core")
             resp["envs"].update(iso["envs"])
             resp["mounts"] = iso["mounts"]
         return resp
@@ -365,17 +370,17 @@ class GPUShareMemoryPlugin(GPUSharePluginBase):
         count = gpu.memory_mib // unit
         return [f"{gpu.index}-{slot:06d}" for slot in range(count)]
 
-    def _allocate_one(self, device: Device, ids: List[str]) -> dict:
-        host = f"/dev/{consts.ELASTIC_GPU_LINK_FMT % (device.hash + '-0')}"
+    def _allocate_one(self, alloc_hash: str, n_units: int) -> dict:
+        host = f"/dev/{consts.ELASTIC_GPU_LINK_FMT % (alloc_hash + '-0')}"
         resp = {
-            "envs": {consts.GPU_ENV_KEY: device.hash},
+            "envs": {consts.GPU_ENV_KEY: alloc_hash},
             "devices": [
                 {"container_path": consts.KFD_PATH, "host_path": consts.KFD_PATH,
                  "permissions": "rw"},
                 {"container_path": "/dev/egpu/gpu0", "host_path": host, "permissions": "rw"},
             ],
         }
-        iso = self._isolation_payload(device, "mem")
+        iso = self._isolation_payload(alloc_hash, "mem")
         resp["envs"].update(iso["envs"])
         if iso["mounts"]:
             resp["mounts"] = iso["mounts"]

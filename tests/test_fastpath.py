@@ -173,3 +173,56 @@ def test_encode_allocate_response_randomized():
             containers.append(c)
         case = {"container_responses": containers}
         assert _fastwire.encode_allocate_response(case) == dp.AllocateResponse.encode(case)
+
+
+def test_digest_allocate_request_matches_device_hash():
+    """(hash, count) digests computed in C++ off the wire must equal
+    Device.new's sorted-join-sha256 identity — including unsorted inputs,
+    empty containers, and the 73k-ID reference-exact 1-MiB scale."""
+    import random
+
+    pytest.importorskip("elastic_gpu_agent_amd._fastwire")
+    from elastic_gpu_agent_amd import _fastwire
+    from elastic_gpu_agent_amd.types import Device
+
+    rng = random.Random(20260914)
+    for _ in range(100):
+        req = {"container_requests": []}
+        expect = []
+        for _ in range(rng.randrange(0, 4)):
+            ids = [f"{rng.randrange(8)}-{rng.randrange(100000):06d}"
+                   for _ in range(rng.randrange(0, 60))]
+            rng.shuffle(ids)
+            req["container_requests"].append({"devicesIDs": ids})
+            expect.append((Device.new(ids).hash, len(ids)))
+        buf = fastpath.encode_allocate_request(req)
+        got = [(h, int(n)) for h, n in _fastwire.digest_allocate_request(buf)]
+        assert got == expect
+        # the wrapper emits handler-ready digest entries
+        wrapped = fastpath.decode_allocate_request_digest(buf)
+        assert [cr["digest"] for cr in wrapped["container_requests"]] == [
+            (h, n) for h, n in got]
+
+    ids = [f"0-{i:06d}" for i in range(73728)]
+    buf = fastpath.encode_allocate_request({"container_requests": [{"devicesIDs": ids}]})
+    (h, n), = _fastwire.digest_allocate_request(buf)
+    assert (h, int(n)) == (Device.new(ids).hash, 73728)
+
+
+def test_allocate_handler_digest_and_ids_paths_agree(tmp_path):
+    """The served fast path (digest) and the dict path (devicesIDs) must
+    produce identical Allocate responses."""
+    from helpers import Harness
+
+    h = Harness(str(tmp_path), gpus=1)
+    try:
+        ids = [f"0-{i:02d}" for i in range(25)]
+        via_ids = h.plugin.core.allocate(
+            {"container_requests": [{"devicesIDs": ids}]}, None)
+        via_digest = h.plugin.core.allocate(
+            fastpath.decode_allocate_request_digest(
+                fastpath.encode_allocate_request(
+                    {"container_requests": [{"devicesIDs": ids}]})), None)
+        assert via_ids == via_digest
+    finally:
+        h.close()

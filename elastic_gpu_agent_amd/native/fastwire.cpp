@@ -256,6 +256,136 @@ py::list digest_allocate_request(py::bytes data) {
   return out;
 }
 
+// ---- GetPreferredAllocation digest (per-GPU counts + on-demand extract) ----
+// At the reference-exact 1-MiB gpu-memory contract kubelet sends the FULL
+// free-ID pool (≈295k IDs, ~3 MB) as available_deviceIDs on every pod
+// admission; decoding that into Python strings plus a Python group-by cost
+// ~300 ms per call. The policy only needs per-GPU availability COUNTS to
+// choose a GPU; the chosen GPU's lexicographically-first `size` IDs are then
+// extracted and emitted as an already-encoded response body without ever
+// materializing Python strings.
+
+struct PreferredContainer {
+  std::vector<std::pair<const uint8_t*, size_t>> available;
+  std::vector<std::pair<const uint8_t*, size_t>> must_include;
+  long long size = 0;
+};
+
+// parse leading "<int>-" GPU prefix of an ID span; -1 when malformed
+long long gpu_prefix(const uint8_t* p, size_t n) {
+  long long v = 0;
+  size_t i = 0;
+  while (i < n && p[i] >= '0' && p[i] <= '9') {
+    v = v * 10 + (p[i] - '0');
+    ++i;
+  }
+  if (i == 0 || i >= n || p[i] != '-') return -1;
+  return v;
+}
+
+void parse_preferred(const uint8_t* p, size_t len,
+                     std::vector<PreferredContainer>& out) {
+  std::vector<std::pair<const uint8_t*, size_t>> outer;
+  field1_spans(p, p + len, outer);
+  out.resize(outer.size());
+  for (size_t i = 0; i < outer.size(); ++i) {
+    Reader r{outer[i].first, outer[i].first + outer[i].second};
+    while (!r.done()) {
+      uint64_t tag = r.varint();
+      uint32_t field = tag >> 3, wire = tag & 7;
+      if ((field == 1 || field == 2) && wire == 2) {
+        uint64_t n = r.varint();
+        if (r.p + n > r.end) throw std::runtime_error("truncated");
+        auto& vec = field == 1 ? out[i].available : out[i].must_include;
+        vec.emplace_back(r.p, (size_t)n);
+        r.p += n;
+      } else if (field == 3 && wire == 0) {
+        out[i].size = (long long)r.varint();
+      } else {
+        r.skip(wire);
+      }
+    }
+  }
+}
+
+py::list preferred_digest(py::bytes data) {
+  char* buf;
+  Py_ssize_t len;
+  PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+  std::vector<PreferredContainer> containers;
+  std::vector<std::vector<std::pair<long long, size_t>>> counts_all;
+  {
+    py::gil_scoped_release rel;
+    parse_preferred((const uint8_t*)buf, (size_t)len, containers);
+    counts_all.resize(containers.size());
+    for (size_t i = 0; i < containers.size(); ++i) {
+      // small flat map: node GPU counts are ≤ tens of entries
+      auto& counts = counts_all[i];
+      for (auto& s : containers[i].available) {
+        long long g = gpu_prefix(s.first, s.second);
+        bool found = false;
+        for (auto& kv : counts)
+          if (kv.first == g) {
+            ++kv.second;
+            found = true;
+            break;
+          }
+        if (!found) counts.emplace_back(g, 1);
+      }
+    }
+  }
+  py::list out;
+  for (size_t i = 0; i < containers.size(); ++i) {
+    py::dict counts;
+    for (auto& kv : counts_all[i])
+      counts[py::int_(kv.first)] = py::int_((long long)kv.second);
+    py::list must;
+    for (auto& s : containers[i].must_include)
+      must.append(py::str((const char*)s.first, s.second));
+    out.append(py::make_tuple(counts, must, containers[i].size));
+  }
+  return out;
+}
+
+// Encoded ContainerPreferredAllocationResponse body (repeated string 1) of
+// the lexicographically-first `n` available IDs of GPU `gpu` in container
+// `index` — same order the Python path produces (groups sorted, then [:n]).
+py::bytes preferred_extract(py::bytes data, size_t index, long long gpu, size_t n) {
+  char* buf;
+  Py_ssize_t len;
+  PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+  std::string out;
+  {
+    py::gil_scoped_release rel;
+    std::vector<PreferredContainer> containers;
+    parse_preferred((const uint8_t*)buf, (size_t)len, containers);
+    if (index >= containers.size()) throw std::runtime_error("bad container index");
+    std::vector<std::pair<const uint8_t*, size_t>> group;
+    for (auto& s : containers[index].available)
+      if (gpu_prefix(s.first, s.second) == gpu) group.push_back(s);
+    auto cmp = [](const std::pair<const uint8_t*, size_t>& a,
+                  const std::pair<const uint8_t*, size_t>& b) {
+      int c = memcmp(a.first, b.first, std::min(a.second, b.second));
+      if (c) return c < 0;
+      return a.second < b.second;
+    };
+    if (n < group.size()) {
+      std::nth_element(group.begin(), group.begin() + n, group.end(), cmp);
+      group.resize(n);
+    }
+    std::sort(group.begin(), group.end(), cmp);
+    size_t est = 0;
+    for (auto& s : group) est += s.second + 6;
+    out.reserve(est);
+    for (auto& s : group) {
+      out.push_back(0x0A);
+      put_varint(out, s.second);
+      out.append((const char*)s.first, s.second);
+    }
+  }
+  return py::bytes(out);
+}
+
 // ---- AllocateResponse encoder (the Allocate hot path's response half) ----
 // Byte-identical to protos.deviceplugin.AllocateResponse.encode (MessageSpec
 // policy: non-repeated strings/bools omitted when empty/false; map entries
@@ -347,4 +477,6 @@ PYBIND11_MODULE(_fastwire, m) {
   m.def("encode_nested_string_lists", &encode_nested_string_lists);
   m.def("encode_allocate_response", &encode_allocate_response);
   m.def("digest_allocate_request", &digest_allocate_request);
+  m.def("preferred_digest", &preferred_digest);
+  m.def("preferred_extract", &preferred_extract);
 }

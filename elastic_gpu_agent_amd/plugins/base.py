@@ -93,7 +93,8 @@ class DevicePluginServer:
             "ListAndWatch": egrpc.unary_stream(list_and_watch),
             "GetPreferredAllocation": timed(
                 "GetPreferredAllocation", p.get_preferred_allocation,
-                dp.PreferredAllocationRequest, dp.PreferredAllocationResponse),
+                _FastSpec(fastpath.decode_preferred_request_digest),
+                _FastSpec(encode=fastpath.encode_preferred_response)),
             "Allocate": timed("Allocate", p.allocate,
                               _FastSpec(fastpath.decode_allocate_request_digest),
                               _FastSpec(encode=fastpath.encode_allocate_response)),

@@ -157,8 +157,41 @@ class GPUSharePluginBase:
                 self.resource_name == consts.RESOURCE_GPU_MEMORY
                 or size <= consts.GPU_PERCENT_EACH_CARD
             )
+            counts = cr.get("counts")
+            if counts is not None and single and not cr.get("must_include_deviceIDs"):
+                # digest fast path: choose the GPU from per-GPU counts alone
+                # (same ranking as prefer_allocation's single-GPU branch:
+                # most-loaded GPU that fits, tie-broken by index), then have
+                # C++ extract that GPU's first `size` IDs as pre-encoded
+                # response bytes — the ~295k-ID pool at 1-MiB memory units
+                # is never materialized in Python.
+                from ..protos import fastpath
+
+                pick = None
+                for g in sorted((g for g in counts if g >= 0),
+                                key=lambda g: (counts[g], g)):
+                    if counts[g] >= size:
+                        pick = g
+                        break
+                if pick is None:
+                    responses.append({"deviceIDs": []})
+                else:
+                    responses.append({
+                        "raw": fastpath.extract_preferred(
+                            cr["_raw"], cr["_index"], pick, size)
+                    })
+                continue
+            avail = cr.get("available_deviceIDs")
+            if avail is None and "_raw" in cr:
+                # digest-decoded request but the generic policy is needed
+                # (must_include present, or a multi-GPU core request)
+                from ..protos import deviceplugin as dp
+
+                full = dp.PreferredAllocationRequest.decode(cr["_raw"])
+                avail = full["container_requests"][cr["_index"]].get(
+                    "available_deviceIDs", [])
             picked = topology.prefer_allocation(
-                cr.get("available_deviceIDs", []),
+                avail or [],
                 cr.get("must_include_deviceIDs", []),
                 size,
                 devices,

@@ -51,6 +51,53 @@ def decode_allocate_request_digest(buf: bytes) -> dict:
     return {"container_requests": out}
 
 
+def decode_preferred_request_digest(buf: bytes) -> dict:
+    """GetPreferredAllocation deserializer for huge ID pools (gpu-memory at
+    1-MiB units: kubelet sends ~295k available IDs per admission). Produces
+    per-GPU availability COUNTS (all the single-GPU pick policy needs) plus
+    the raw buffer for on-demand extraction of the chosen GPU's IDs —
+    never materializing the pool as Python strings."""
+    if _fastwire is not None and hasattr(_fastwire, "preferred_digest"):
+        crs = []
+        for i, (counts, must, size) in enumerate(_fastwire.preferred_digest(buf)):
+            crs.append(
+                {
+                    "counts": {int(g): int(n) for g, n in counts.items()},
+                    "must_include_deviceIDs": list(must),
+                    "allocation_size": int(size),
+                    "_raw": buf,
+                    "_index": i,
+                }
+            )
+        return {"container_requests": crs}
+    return dp.PreferredAllocationRequest.decode(buf)
+
+
+def extract_preferred(raw: bytes, index: int, gpu: int, n: int) -> bytes:
+    """Encoded ContainerPreferredAllocationResponse body: the lexicographically
+    first ``n`` available IDs of ``gpu`` in container ``index`` of ``raw``."""
+    return _fastwire.preferred_extract(raw, index, gpu, n)
+
+
+def encode_preferred_response(resp: dict) -> bytes:
+    """PreferredAllocationResponse encoder accepting either materialized
+    ``deviceIDs`` lists or pre-encoded ``raw`` container bodies (the digest
+    path's zero-materialization output)."""
+    out = bytearray()
+    for cr in resp.get("container_responses", []):
+        body = cr.get("raw")
+        if body is None:
+            ids = cr.get("deviceIDs", [])
+            if _fastwire is not None:
+                body = _fastwire.encode_string_list(ids)
+            else:
+                body = b"".join(
+                    b"\x0a" + encode_varint(len(i.encode())) + i.encode() for i in ids
+                )
+        out += b"\x0a" + encode_varint(len(body)) + body
+    return bytes(out)
+
+
 def encode_allocate_request(req: dict) -> bytes:
     """Client-side fast path (bench / tests; kubelet's own Go encoder plays
     this role in production)."""

@@ -226,3 +226,70 @@ def test_allocate_handler_digest_and_ids_paths_agree(tmp_path):
         assert via_ids == via_digest
     finally:
         h.close()
+
+
+def test_preferred_digest_path_matches_generic(tmp_path):
+    """GetPreferredAllocation through the digest fast path (per-GPU counts +
+    C++ extraction, no Python materialization of the ID pool) must be
+    byte-identical to the generic prefer_allocation path — including the
+    not-enough-room, multi-container, and must_include-fallback cases."""
+    import random
+
+    pytest.importorskip("elastic_gpu_agent_amd._fastwire")
+    from helpers import Harness
+
+    h = Harness(str(tmp_path), gpus=2, mem_unit_mib=512)
+    rng = random.Random(42)
+    try:
+        for plugin in (h.plugin.memory, h.plugin.core):
+            for trial in range(30):
+                crs = []
+                for _ in range(rng.randrange(1, 3)):
+                    ids = [f"{rng.randrange(2)}-{rng.randrange(10000):06d}"
+                           for _ in range(rng.randrange(0, 400))]
+                    rng.shuffle(ids)
+                    cr = {"available_deviceIDs": list(dict.fromkeys(ids)),
+                          "allocation_size": rng.choice([0, 1, 13, 97, 150, 999])}
+                    if rng.random() < 0.2 and cr["available_deviceIDs"]:
+                        cr["must_include_deviceIDs"] = [cr["available_deviceIDs"][0]]
+                    crs.append(cr)
+                req = {"container_requests": crs}
+                buf = dp.PreferredAllocationRequest.encode(req)
+                digest_resp = plugin.get_preferred_allocation(
+                    fastpath.decode_preferred_request_digest(buf), None)
+                generic_resp = plugin.get_preferred_allocation(
+                    dp.PreferredAllocationRequest.decode(buf), None)
+                assert (fastpath.encode_preferred_response(digest_resp)
+                        == dp.PreferredAllocationResponse.encode(generic_resp)), (
+                    plugin.resource_name, trial, req)
+    finally:
+        h.close()
+
+
+def test_preferred_digest_large_pool_fast(tmp_path):
+    """295k-ID pool (reference-exact 1-MiB units on 288 GiB): the digest path
+    must answer well under the generic path's ~300 ms."""
+    import time
+
+    pytest.importorskip("elastic_gpu_agent_amd._fastwire")
+    from helpers import Harness
+
+    h = Harness(str(tmp_path), gpus=2, mem_unit_mib=1)
+    try:
+        avail = [f"{g}-{i:06d}" for g in range(2) for i in range(147456)]
+        buf = dp.PreferredAllocationRequest.encode(
+            {"container_requests": [{"available_deviceIDs": avail,
+                                     "allocation_size": 73728}]})
+        t0 = time.perf_counter()
+        resp = h.plugin.memory.get_preferred_allocation(
+            fastpath.decode_preferred_request_digest(buf), None)
+        out = fastpath.encode_preferred_response(resp)
+        elapsed = time.perf_counter() - t0
+        ids = dp.PreferredAllocationResponse.decode(out)[
+            "container_responses"][0]["deviceIDs"]
+        assert len(ids) == 73728
+        assert len({i.split("-")[0] for i in ids}) == 1  # single GPU
+        assert ids == sorted(ids)
+        assert elapsed < 2.0, f"digest path too slow: {elapsed:.2f}s"
+    finally:
+        h.close()

@@ -176,3 +176,56 @@ def test_locator_lazy_reconnect(tmp_path):
     finally:
         loc.close()
         server2.stop()
+
+
+def test_preferred_allocation_over_the_wire(h):
+    """GetPreferredAllocation through the served socket — exercising the
+    digest deserializer + raw-body serializer exactly as kubelet would —
+    from BOTH client stacks (egrpc and grpcio)."""
+    from helpers import GrpcioPluginClient
+
+    h.plugin.core_server.serve()
+    h.plugin.memory_server.serve()
+    try:
+        for mk in (PluginClient, GrpcioPluginClient):
+            core = mk(h.plugin.core_server.socket_path)
+            mem = mk(h.plugin.memory_server.socket_path)
+            try:
+                avail = [f"{g}-{i:02d}" for g in range(2) for i in range(100)]
+                resp = core.preferred({"container_requests": [
+                    {"available_deviceIDs": avail, "allocation_size": 30}]})
+                ids = resp["container_responses"][0]["deviceIDs"]
+                assert len(ids) == 30
+                assert len({i.split("-")[0] for i in ids}) == 1  # single GPU
+
+                # multi-GPU core request (falls back to the generic policy)
+                resp = core.preferred({"container_requests": [
+                    {"available_deviceIDs": avail, "allocation_size": 200}]})
+                assert len(resp["container_responses"][0]["deviceIDs"]) == 200
+
+                # memory request with a larger pool
+                mem_avail = [f"{g}-{i:06d}" for g in range(2) for i in range(5000)]
+                resp = mem.preferred({"container_requests": [
+                    {"available_deviceIDs": mem_avail, "allocation_size": 4096},
+                    {"available_deviceIDs": [], "allocation_size": 1},
+                ]})
+                ids = resp["container_responses"][0]["deviceIDs"]
+                assert len(ids) == 4096
+                assert ids == sorted(ids)
+                assert len({i.split("-")[0] for i in ids}) == 1
+                assert resp["container_responses"][1].get("deviceIDs", []) == []
+
+                # must_include honored through the wire (generic fallback)
+                resp = core.preferred({"container_requests": [
+                    {"available_deviceIDs": avail,
+                     "must_include_deviceIDs": ["1-07"],
+                     "allocation_size": 10}]})
+                ids = resp["container_responses"][0]["deviceIDs"]
+                assert "1-07" in ids and len(ids) == 10
+                assert all(i.startswith("1-") for i in ids)
+            finally:
+                core.close()
+                mem.close()
+    finally:
+        h.plugin.core_server.stop()
+        h.plugin.memory_server.stop()

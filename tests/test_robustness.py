@@ -190,3 +190,34 @@ def test_rpc_latency_recorded_through_server(tmp_path):
     assert after == before + 1
     client.close()
     h.close()
+
+
+def test_malformed_protobuf_bodies_fail_cleanly(tmp_path):
+    """Garbage protobuf in a request body (through the C++ digest
+    deserializers) must surface as a per-RPC error status — never kill the
+    connection or the server."""
+    from helpers import Harness
+    from elastic_gpu_agent_amd import egrpc
+    from elastic_gpu_agent_amd.protos import deviceplugin as dp
+
+    h = Harness(str(tmp_path), gpus=1)
+    h.plugin.core_server.serve()
+    try:
+        ch = egrpc.Channel(h.plugin.core_server.socket_path)
+        for method in (dp.METHOD_ALLOCATE, dp.METHOD_GET_PREFERRED_ALLOCATION,
+                       dp.METHOD_PRE_START_CONTAINER):
+            call = ch.unary_unary(method, request_serializer=lambda x: x,
+                                  response_deserializer=lambda b: b)
+            for bad in (b"\xff\xff\xff\xff", b"\x0a\xff\x01garbage", b"\x0a",
+                        b"\x0a\x05\x0a\xff\xff\xff\xff"):
+                with pytest.raises(egrpc.EgrpcError):
+                    call(bad)
+        # connection and server both survive the abuse
+        opts = ch.unary_unary(
+            dp.METHOD_GET_OPTIONS, request_serializer=dp.Empty.encode,
+            response_deserializer=dp.DevicePluginOptions.decode)({})
+        assert opts["pre_start_required"] is True
+        ch.close()
+    finally:
+        h.plugin.core_server.stop()
+        h.close()

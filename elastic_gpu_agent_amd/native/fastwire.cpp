@@ -256,6 +256,44 @@ py::list digest_allocate_request(py::bytes data) {
   return out;
 }
 
+// PreStartContainerRequest digest: the handler needs the SORTED ID list
+// (persisted in the reference's on-disk record format) plus the device-set
+// hash; one C++ pass sorts the spans, hashes, then materializes the sorted
+// Python strings exactly once.
+py::tuple decode_prestart_digest(py::bytes data) {
+  char* buf;
+  Py_ssize_t len;
+  PyBytes_AsStringAndSize(data.ptr(), &buf, &len);
+  std::vector<std::pair<const uint8_t*, size_t>> ids;
+  char hex[9];
+  {
+    py::gil_scoped_release rel;
+    field1_spans((const uint8_t*)buf, (const uint8_t*)buf + len, ids);
+    auto cmp = [](const std::pair<const uint8_t*, size_t>& a,
+                  const std::pair<const uint8_t*, size_t>& b) {
+      int c = memcmp(a.first, b.first, std::min(a.second, b.second));
+      if (c) return c < 0;
+      return a.second < b.second;
+    };
+    if (!std::is_sorted(ids.begin(), ids.end(), cmp))
+      std::sort(ids.begin(), ids.end(), cmp);
+    EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+    EVP_DigestInit_ex(ctx, EVP_sha256(), nullptr);
+    for (size_t i = 0; i < ids.size(); ++i) {
+      if (i) EVP_DigestUpdate(ctx, ":", 1);
+      EVP_DigestUpdate(ctx, ids[i].first, ids[i].second);
+    }
+    unsigned char md[32];
+    unsigned int mdlen = 0;
+    EVP_DigestFinal_ex(ctx, md, &mdlen);
+    EVP_MD_CTX_free(ctx);
+    snprintf(hex, sizeof hex, "%02x%02x%02x%02x", md[0], md[1], md[2], md[3]);
+  }
+  py::list out;
+  for (auto& s : ids) out.append(py::str((const char*)s.first, s.second));
+  return py::make_tuple(out, py::str(hex, 8));
+}
+
 // ---- GetPreferredAllocation digest (per-GPU counts + on-demand extract) ----
 // At the reference-exact 1-MiB gpu-memory contract kubelet sends the FULL
 // free-ID pool (≈295k IDs, ~3 MB) as available_deviceIDs on every pod
@@ -477,6 +515,7 @@ PYBIND11_MODULE(_fastwire, m) {
   m.def("encode_nested_string_lists", &encode_nested_string_lists);
   m.def("encode_allocate_response", &encode_allocate_response);
   m.def("digest_allocate_request", &digest_allocate_request);
+  m.def("decode_prestart_digest", &decode_prestart_digest);
   m.def("preferred_digest", &preferred_digest);
   m.def("preferred_extract", &preferred_extract);
 }

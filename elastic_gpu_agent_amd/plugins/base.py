@@ -100,7 +100,7 @@ class DevicePluginServer:
                               _FastSpec(encode=fastpath.encode_allocate_response)),
             "PreStartContainer": timed(
                 "PreStartContainer", p.pre_start_container,
-                _FastSpec(fastpath.decode_prestart_request),
+                _FastSpec(fastpath.decode_prestart_request_digest),
                 dp.PreStartContainerResponse),
         }
 

@@ -248,7 +248,11 @@ class GPUSharePluginBase:
 
     def pre_start_container(self, request, context) -> dict:
         ids = request.get("devicesIDs", [])
-        device = Device.new(ids, self.resource_name)
+        h8 = request.get("device_hash")
+        if h8 is not None:  # digest deserializer: ids pre-sorted, hash ready
+            device = Device(hash=h8, list=tuple(ids), resource_name=self.resource_name)
+        else:
+            device = Device.new(ids, self.resource_name)
         from .. import egrpc
 
         try:

@@ -129,6 +129,16 @@ def decode_prestart_request(buf: bytes) -> dict:
     return dp.PreStartContainerRequest.decode(buf)
 
 
+def decode_prestart_request_digest(buf: bytes) -> dict:
+    """PreStart deserializer: sorted ID list (persisted in the reference's
+    on-disk record) plus the device-set hash, both from one C++ pass —
+    skipping the Python re-sort + join + sha256 at 1-MiB unit scale."""
+    if _fastwire is not None and hasattr(_fastwire, "decode_prestart_digest"):
+        ids, h = _fastwire.decode_prestart_digest(buf)
+        return {"devicesIDs": ids, "device_hash": h}
+    return decode_prestart_request(buf)
+
+
 # Precomputable per-GPU Device suffix: health + topology are identical for
 # every fake device of a GPU.
 def device_suffix(health: str, numa_node: int) -> bytes:

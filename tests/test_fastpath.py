@@ -293,3 +293,46 @@ def test_preferred_digest_large_pool_fast(tmp_path):
         assert elapsed < 2.0, f"digest path too slow: {elapsed:.2f}s"
     finally:
         h.close()
+
+
+def test_prestart_digest_matches_device_new(tmp_path):
+    """decode_prestart_request_digest must deliver exactly Device.new's
+    sorted list and hash, and the PreStart handler must persist the same
+    record either way."""
+    import random
+
+    pytest.importorskip("elastic_gpu_agent_amd._fastwire")
+    from elastic_gpu_agent_amd.types import Device
+
+    rng = random.Random(99)
+    for _ in range(50):
+        ids = [f"{rng.randrange(4)}-{rng.randrange(100000):06d}"
+               for _ in range(rng.randrange(0, 80))]
+        rng.shuffle(ids)
+        buf = fastpath.encode_prestart_request({"devicesIDs": ids})
+        req = fastpath.decode_prestart_request_digest(buf)
+        d = Device.new(ids)
+        assert req["devicesIDs"] == list(d.list)
+        assert req["device_hash"] == d.hash
+
+    # handler round-trip: digest-form request persists an identical record
+    from elastic_gpu_agent_amd import consts
+    from elastic_gpu_agent_amd.types import PodContainer
+    from helpers import Harness
+
+    h = Harness(str(tmp_path), gpus=1)
+    try:
+        ids = [f"0-{i:02d}" for i in range(30)]
+        d = Device.new(ids, consts.RESOURCE_GPU_CORE)
+        h.core_locator.assign(d.hash, PodContainer("ns", "pd", "main"))
+        h.add_assumed_pod("ns", "pd", "main", "0")
+        buf = fastpath.encode_prestart_request({"devicesIDs": ids})
+        h.plugin.core.pre_start_container(
+            fastpath.decode_prestart_request_digest(buf), None)
+        pi = h.storage.load("ns", "pd")
+        stored = pi.container_device_map["main"]
+        assert stored.hash == d.hash
+        assert stored.list == d.list
+        assert stored.resource_name == consts.RESOURCE_GPU_CORE
+    finally:
+        h.close()

@@ -1,0 +1,141 @@
+"""Property-based invariants (hypothesis) for the allocation-critical logic.
+
+Each property encodes a safety rule the randomized scenario tests rely on
+implicitly; hypothesis searches the input space and shrinks failures to
+minimal counterexamples. `derandomize=True` keeps CI deterministic.
+"""
+from __future__ import annotations
+
+from hypothesis import given, settings, strategies as st
+
+from elastic_gpu_agent_amd import consts
+from elastic_gpu_agent_amd.isolation.cumask import (
+    cu_count_for_percent,
+    mask_hex,
+    mask_words_from_cus,
+    parse_mask_hex,
+    popcount,
+    xcd_round_robin_cus,
+)
+from elastic_gpu_agent_amd.topology import prefer_allocation
+from elastic_gpu_agent_amd.types import Device
+
+SETTINGS = settings(max_examples=200, derandomize=True, deadline=None)
+
+
+# ---- device-set identity ----
+
+ids_strategy = st.lists(
+    st.from_regex(r"[0-7]-[0-9]{2,6}", fullmatch=True), min_size=0, max_size=40
+)
+
+
+@SETTINGS
+@given(ids=ids_strategy)
+def test_device_hash_order_invariant(ids):
+    """The device-set hash must not depend on request order (kubelet gives
+    no ordering guarantee) and must match a fresh sort."""
+    import random
+
+    shuffled = list(ids)
+    random.Random(0).shuffle(shuffled)
+    a = Device.new(ids, consts.RESOURCE_GPU_CORE)
+    b = Device.new(shuffled, consts.RESOURCE_GPU_CORE)
+    assert a.hash == b.hash
+    assert a.list == b.list == tuple(sorted(ids))
+
+
+@SETTINGS
+@given(ids=ids_strategy)
+def test_digest_matches_device_hash(ids):
+    """C++ wire digest == Python Device identity for arbitrary ID sets."""
+    try:
+        from elastic_gpu_agent_amd import _fastwire
+    except ImportError:
+        return
+    from elastic_gpu_agent_amd.protos import fastpath
+
+    buf = fastpath.encode_allocate_request(
+        {"container_requests": [{"devicesIDs": ids}]}
+    )
+    (h, n), = _fastwire.digest_allocate_request(buf)
+    d = Device.new(ids)
+    assert (h, int(n)) == (d.hash, len(ids))
+
+
+# ---- CU masks ----
+
+
+@SETTINGS
+@given(percent=st.integers(min_value=0, max_value=100))
+def test_cu_count_pair_granular_and_monotone(percent):
+    n = cu_count_for_percent(percent, consts.GFX950_CU_COUNT)
+    assert n % 2 == 0  # ROCr CU masks are pair-granular
+    assert 0 <= n <= consts.GFX950_CU_COUNT
+    if percent > 0:
+        assert n >= 2  # any nonzero ask gets at least one pair
+        assert n >= cu_count_for_percent(percent - 1, consts.GFX950_CU_COUNT)
+
+
+@SETTINGS
+@given(n=st.integers(min_value=0, max_value=256))
+def test_xcd_round_robin_mask_roundtrip(n):
+    cus = xcd_round_robin_cus(n if n % 2 == 0 else n + 1,
+                              consts.GFX950_CU_COUNT, consts.GFX950_XCD_COUNT)
+    words = mask_words_from_cus(cus, consts.GFX950_CU_COUNT)
+    hexmask = mask_hex(words)
+    assert popcount(parse_mask_hex(hexmask)) == len(cus)
+    assert len(set(cus)) == len(cus)  # no CU twice
+    # spread: no XCD holds more than ceil(len/xcds)+pair worth extra
+    per_xcd = consts.GFX950_CU_COUNT // consts.GFX950_XCD_COUNT
+    by_xcd = {}
+    for cu in cus:
+        by_xcd.setdefault(cu // per_xcd, 0)
+        by_xcd[cu // per_xcd] += 1
+    if cus:
+        assert max(by_xcd.values()) - min(by_xcd.values() or [0]) <= 2
+
+
+# ---- preferred allocation ----
+
+avail_strategy = st.lists(
+    st.tuples(st.integers(min_value=0, max_value=3),
+              st.integers(min_value=0, max_value=99)),
+    min_size=0, max_size=120, unique=True,
+).map(lambda pairs: [f"{g}-{s:02d}" for g, s in pairs])
+
+
+def _devs(n=4):
+    from elastic_gpu_agent_amd.types import GPUDevice
+
+    return [
+        GPUDevice(uuid=f"u{i}", index=i, memory_bytes=288 << 30,
+                  drm_render_minor=128 + i, cu_count=256, xcd_count=8,
+                  numa_node=i // 2, xgmi_peers=[j for j in range(n) if j != i])
+        for i in range(n)
+    ]
+
+
+@SETTINGS
+@given(avail=avail_strategy, size=st.integers(min_value=0, max_value=130),
+       single=st.booleans())
+def test_prefer_allocation_invariants(avail, size, single):
+    picked = prefer_allocation(avail, [], size, _devs(), single_gpu=single)
+    assert len(picked) == len(set(picked))  # no duplicates
+    assert set(picked) <= set(avail)  # only offered IDs
+    assert len(picked) <= size or size == 0
+    if single and picked:
+        assert len({p.split("-")[0] for p in picked}) == 1  # one GPU only
+    if not single:
+        # full pick whenever enough IDs exist
+        assert len(picked) == min(size, len(avail))
+
+
+@SETTINGS
+@given(avail=avail_strategy, size=st.integers(min_value=1, max_value=60))
+def test_prefer_allocation_must_include_honored(avail, size):
+    if not avail:
+        return
+    must = [avail[0]]
+    picked = prefer_allocation(avail, must, size, _devs())
+    assert must[0] in picked

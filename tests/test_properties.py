@@ -139,3 +139,61 @@ def test_prefer_allocation_must_include_honored(avail, size):
     must = [avail[0]]
     picked = prefer_allocation(avail, must, size, _devs())
     assert must[0] in picked
+
+
+# ---- wire codecs ----
+
+header_strategy = st.lists(
+    st.tuples(
+        st.text(alphabet=st.characters(min_codepoint=0x21, max_codepoint=0x7E),
+                min_size=1, max_size=24).map(str.lower),
+        st.text(alphabet=st.characters(min_codepoint=0x20, max_codepoint=0x7E),
+                max_size=48),
+    ),
+    min_size=0, max_size=12,
+)
+
+
+@SETTINGS
+@given(headers=header_strategy)
+def test_hpack_roundtrip(headers):
+    """Our HPACK encoder's output must decode back to the same header list
+    (both the Python decoder and the C++ one used by _etransport)."""
+    from elastic_gpu_agent_amd.egrpc import hpack
+
+    hdrs = [(k.encode(), v.encode()) for k, v in headers]
+    block = hpack.encode_headers(hdrs)
+    assert hpack.Decoder().decode(block) == hdrs
+    try:
+        from elastic_gpu_agent_amd import _etransport
+    except ImportError:
+        return
+    if hasattr(_etransport, "HpackTester"):
+        decoded = _etransport.HpackTester().decode(block)
+        norm = [
+            (k.encode() if isinstance(k, str) else bytes(k),
+             v.encode() if isinstance(v, str) else bytes(v))
+            for k, v in decoded
+        ]
+        assert norm == hdrs
+
+
+@SETTINGS
+@given(
+    crs=st.lists(
+        st.lists(st.from_regex(r"[0-7]-[0-9]{2,6}", fullmatch=True),
+                 min_size=0, max_size=20),
+        min_size=0, max_size=4,
+    )
+)
+def test_allocate_request_wire_roundtrip(crs):
+    """encode → decode identity for AllocateRequest through both codecs."""
+    from elastic_gpu_agent_amd.protos import deviceplugin as dp, fastpath
+
+    req = {"container_requests": [{"devicesIDs": ids} for ids in crs]}
+    buf = fastpath.encode_allocate_request(req)
+    assert buf == dp.AllocateRequest.encode(req)
+    dec = fastpath.decode_allocate_request(buf)
+    got = [list(cr.get("devicesIDs", [])) for cr in dec.get("container_requests", [])]
+    # proto3 cannot distinguish absent vs empty repeated containers
+    assert [g for g in got if g] == [c for c in crs if c]

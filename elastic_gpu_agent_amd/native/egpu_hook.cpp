@@ -322,7 +322,7 @@ int inject(long pid, const std::vector<DeviceNode>& nodes, bool dryrun,
 // real kernel verifier / a scratch cgroup):
 //   egpu-hook devfilter-load <config.json>
 //   egpu-hook devfilter-attach <cgroup_dir> <config.json>
-int devfilter_cmd(int argc, char** argv) {
+int devfilter_cmd(char** argv) {
   bool do_attach = strcmp(argv[1], "devfilter-attach") == 0;
   const char* cfg_path = do_attach ? argv[3] : argv[2];
   std::string config = slurp_file(cfg_path);
@@ -353,7 +353,7 @@ int devfilter_cmd(int argc, char** argv) {
 int main(int argc, char** argv) {
   if (argc > 2 && (strcmp(argv[1], "devfilter-load") == 0 ||
                    (argc > 3 && strcmp(argv[1], "devfilter-attach") == 0))) {
-    return devfilter_cmd(argc, argv);
+    return devfilter_cmd(argv);
   }
   // accept NVIDIA-hook-style lifecycle argument; only prestart acts
   if (argc > 1 && strcmp(argv[1], "prestart") != 0 && strcmp(argv[1], "createRuntime") != 0) {
